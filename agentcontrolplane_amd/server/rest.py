@@ -1,0 +1,345 @@
+"""North-bound REST API.
+
+Parity with acp/internal/server/server.go (1,545 LoC; gin on :8082):
+
+    GET    /status
+    GET    /v1/tasks            GET /v1/tasks/{id}        POST /v1/tasks
+    GET    /v1/agents           GET /v1/agents/{name}
+    POST   /v1/agents           PUT /v1/agents/{name}     DELETE /v1/agents/{name}
+    POST   /v1/beta3/events
+
+createTask semantics (server.go:1274-1382): strict JSON (unknown fields are
+400), exactly one of userMessage/contextWindow, 404 when the agent is
+missing, generated name ``<agent>-task-<rand8>``, 201 with the sanitized
+task.  createAgent (server.go:219-439) creates the Secret + LLM + MCP
+servers from an inline config.  The v1beta3 event endpoint auto-creates
+Secret + ContactChannel ``v1beta3-channel-<id>`` + a Task labeled
+``acp.humanlayer.dev/v1beta3=true`` (server.go:1384-1545).
+
+Implemented as a FastAPI app (uvicorn-servable); handlers call the store
+directly, the reconcilers react through watches — exactly the reference's
+relationship between its gin server and the controller caches.
+
+Extra (no reference counterpart): GET /v1/approvals + POST
+/v1/approvals/{id} expose the in-process human-approval queue that replaces
+the HumanLayer SaaS, and /metrics exposes engine/controller counters.
+"""
+from __future__ import annotations
+
+import json
+from typing import Any, Dict, List, Optional
+
+from fastapi import FastAPI, Request, Response
+from fastapi.responses import JSONResponse
+
+from ..api.types import (
+    AGENT,
+    CONTACT_CHANNEL,
+    LLM,
+    SECRET,
+    TASK,
+    make_resource,
+)
+from ..api.validation import (
+    ValidationError,
+    generate_k8s_random_string,
+    validate_task_message_input,
+)
+from ..humanlayer.client import APPROVAL, respond_to_approval, respond_to_contact
+
+
+def _sanitize_task(task: Dict[str, Any]) -> Dict[str, Any]:
+    """server.go:618-625 — the API view of a Task."""
+    return {
+        "name": task["metadata"]["name"],
+        "namespace": task["metadata"].get("namespace", "default"),
+        "agentName": task.get("spec", {}).get("agentRef", {}).get("name", ""),
+        "userMessage": task.get("spec", {}).get("userMessage", ""),
+        "phase": task.get("status", {}).get("phase", ""),
+        "status": task.get("status", {}).get("status", ""),
+        "statusDetail": task.get("status", {}).get("statusDetail", ""),
+        "output": task.get("status", {}).get("output", ""),
+        "contextWindow": task.get("status", {}).get("contextWindow", []),
+        "createdAt": task["metadata"].get("creationTimestamp", ""),
+    }
+
+
+def build_app(store, manager=None, engine=None) -> FastAPI:
+    app = FastAPI(title="acp-amd", version="0.1.0")
+
+    @app.get("/status")
+    def get_status():
+        return {"status": "ok", "resources": store.stats()}
+
+    # ------------------------------------------------------------------ tasks
+
+    @app.get("/v1/tasks")
+    def list_tasks(namespace: str = "default"):
+        return [_sanitize_task(t) for t in store.list(TASK, namespace)]
+
+    @app.get("/v1/tasks/{task_id}")
+    def get_task(task_id: str, namespace: str = "default"):
+        t = store.get(TASK, task_id, namespace)
+        if t is None:
+            return JSONResponse({"error": "Task not found"}, status_code=404)
+        return _sanitize_task(t)
+
+    @app.post("/v1/tasks")
+    async def create_task(request: Request):
+        raw = await request.body()
+        try:
+            body = json.loads(raw)
+        except json.JSONDecodeError as e:
+            return JSONResponse({"error": f"Invalid JSON format: {e}"}, status_code=400)
+        allowed = {"agentName", "userMessage", "contextWindow", "namespace", "contactChannelRef"}
+        unknown = set(body) - allowed
+        if unknown:
+            return JSONResponse(
+                {"error": f"Unknown field in request: {sorted(unknown)}"}, status_code=400
+            )
+        agent_name = body.get("agentName", "")
+        if not agent_name:
+            return JSONResponse({"error": "agentName is required"}, status_code=400)
+        try:
+            validate_task_message_input(
+                body.get("userMessage", ""), body.get("contextWindow")
+            )
+        except ValidationError as e:
+            return JSONResponse({"error": str(e)}, status_code=400)
+        ns = body.get("namespace", "default")
+        store.ensure_namespace(ns)
+        if store.get(AGENT, agent_name, ns) is None:
+            return JSONResponse({"error": "Agent not found"}, status_code=404)
+        name = f"{agent_name}-task-{generate_k8s_random_string(8)}"
+        spec: Dict[str, Any] = {"agentRef": {"name": agent_name}}
+        if body.get("userMessage"):
+            spec["userMessage"] = body["userMessage"]
+        if body.get("contextWindow"):
+            spec["contextWindow"] = body["contextWindow"]
+        if body.get("contactChannelRef"):
+            spec["contactChannelRef"] = {"name": body["contactChannelRef"]}
+        task = store.create(make_resource(TASK, name, ns, spec))
+        return JSONResponse(_sanitize_task(task), status_code=201)
+
+    # ----------------------------------------------------------------- agents
+
+    @app.get("/v1/agents")
+    def list_agents(namespace: str = "default"):
+        return [
+            {
+                "name": a["metadata"]["name"],
+                "ready": a.get("status", {}).get("ready", False),
+                "status": a.get("status", {}).get("status", ""),
+                "systemPrompt": a.get("spec", {}).get("system", ""),
+            }
+            for a in store.list(AGENT, namespace)
+        ]
+
+    @app.get("/v1/agents/{name}")
+    def get_agent(name: str, namespace: str = "default"):
+        a = store.get(AGENT, name, namespace)
+        if a is None:
+            return JSONResponse({"error": "Agent not found"}, status_code=404)
+        return {
+            "name": a["metadata"]["name"],
+            "ready": a.get("status", {}).get("ready", False),
+            "status": a.get("status", {}).get("status", ""),
+            "statusDetail": a.get("status", {}).get("statusDetail", ""),
+            "systemPrompt": a.get("spec", {}).get("system", ""),
+            "mcpServers": a.get("spec", {}).get("mcpServers", []),
+        }
+
+    @app.post("/v1/agents")
+    async def create_agent(request: Request):
+        """server.go:219-439 — inline LLM + secret + MCP server creation."""
+        raw = await request.body()
+        try:
+            body = json.loads(raw)
+        except json.JSONDecodeError as e:
+            return JSONResponse({"error": f"Invalid request body: {e}"}, status_code=400)
+        name = body.get("name", "")
+        system_prompt = body.get("systemPrompt", "")
+        llm_cfg = body.get("llm", {}) or {}
+        ns = body.get("namespace", "default")
+        if not name or not system_prompt:
+            return JSONResponse({"error": "name and systemPrompt are required"}, status_code=400)
+        if llm_cfg and llm_cfg.get("provider") not in (
+            "openai", "anthropic", "mistral", "google", "vertex", "mock", "local",
+        ):
+            return JSONResponse(
+                {"error": "invalid llm provider: " + str(llm_cfg.get("provider"))},
+                status_code=400,
+            )
+        if store.get(AGENT, name, ns) is not None:
+            return JSONResponse({"error": "Agent already exists"}, status_code=409)
+
+        llm_name = body.get("llmRef") or (llm_cfg.get("name") or f"{name}-llm")
+        if llm_cfg:
+            llm_spec: Dict[str, Any] = {
+                "provider": llm_cfg.get("provider", ""),
+                "parameters": {"model": llm_cfg.get("model", "")},
+            }
+            if llm_cfg.get("apiKey"):
+                secret_name = f"{llm_name}-credentials"
+                if store.get(SECRET, secret_name, ns) is None:
+                    store.create(
+                        make_resource(
+                            SECRET, secret_name, ns,
+                            {"data": {"api-key": llm_cfg["apiKey"]}},
+                            api_version="v1",
+                        )
+                    )
+                llm_spec["apiKeyFrom"] = {
+                    "secretKeyRef": {"name": secret_name, "key": "api-key"}
+                }
+            if store.get(LLM, llm_name, ns) is None:
+                store.create(make_resource(LLM, llm_name, ns, llm_spec))
+        elif store.get(LLM, llm_name, ns) is None:
+            return JSONResponse({"error": "LLM not found"}, status_code=404)
+
+        mcp_refs: List[Dict[str, str]] = []
+        for mname, mcfg in (body.get("mcpServers", {}) or {}).items():
+            full = f"{name}-{mname}"
+            if store.get("MCPServer", full, ns) is None:
+                store.create(
+                    make_resource(
+                        "MCPServer",
+                        full,
+                        ns,
+                        {
+                            "transport": mcfg.get("transport", "stdio"),
+                            "command": mcfg.get("command", ""),
+                            "args": mcfg.get("args", []),
+                            "env": mcfg.get("env", []),
+                        },
+                    )
+                )
+            mcp_refs.append({"name": full})
+
+        spec = {"llmRef": {"name": llm_name}, "system": system_prompt}
+        if mcp_refs:
+            spec["mcpServers"] = mcp_refs
+        if body.get("subAgents"):
+            spec["subAgents"] = [{"name": s} for s in body["subAgents"]]
+        agent = store.create(make_resource(AGENT, name, ns, spec))
+        return JSONResponse(
+            {"name": agent["metadata"]["name"], "status": "Pending"}, status_code=201
+        )
+
+    @app.put("/v1/agents/{name}")
+    async def update_agent(name: str, request: Request, namespace: str = "default"):
+        a = store.get(AGENT, name, namespace)
+        if a is None:
+            return JSONResponse({"error": "Agent not found"}, status_code=404)
+        body = json.loads(await request.body())
+        if body.get("systemPrompt"):
+            a["spec"]["system"] = body["systemPrompt"]
+        if body.get("subAgents") is not None:
+            a["spec"]["subAgents"] = [{"name": s} for s in body["subAgents"]]
+        store.update(a)
+        return {"name": name, "status": "updated"}
+
+    @app.delete("/v1/agents/{name}")
+    def delete_agent(name: str, namespace: str = "default"):
+        if not store.delete(AGENT, name, namespace):
+            return JSONResponse({"error": "Agent not found"}, status_code=404)
+        return Response(status_code=204)
+
+    # ---------------------------------------------------------------- v1beta3
+
+    @app.post("/v1/beta3/events")
+    async def v1beta3_event(request: Request):
+        """server.go:1384-1545 — Secret + ContactChannel + labeled Task."""
+        body = json.loads(await request.body())
+        if body.get("type") != "agent_email.received":
+            return JSONResponse({"error": "unsupported event type"}, status_code=400)
+        ev = body.get("data", {}) or {}
+        event_id = ev.get("eventId") or generate_k8s_random_string(8)
+        ns = body.get("namespace", "default")
+        agent_name = ev.get("agentName", "")
+        if not agent_name or store.get(AGENT, agent_name, ns) is None:
+            return JSONResponse({"error": "Agent not found"}, status_code=404)
+        secret_name = f"v1beta3-secret-{event_id}"
+        if store.get(SECRET, secret_name, ns) is None:
+            store.create(
+                make_resource(
+                    SECRET, secret_name, ns,
+                    {"data": {"api-key": ev.get("apiKey", "hl-v1beta3")}},
+                    api_version="v1",
+                )
+            )
+        channel_name = f"v1beta3-channel-{event_id}"
+        if store.get(CONTACT_CHANNEL, channel_name, ns) is None:
+            store.create(
+                make_resource(
+                    CONTACT_CHANNEL,
+                    channel_name,
+                    ns,
+                    {
+                        "type": "email",
+                        "apiKeyFrom": {"secretKeyRef": {"name": secret_name, "key": "api-key"}},
+                        "email": {"address": ev.get("fromAddress", "user@example.com")},
+                    },
+                )
+            )
+        task_name = f"{agent_name}-task-{generate_k8s_random_string(8)}"
+        task = make_resource(
+            TASK,
+            task_name,
+            ns,
+            {
+                "agentRef": {"name": agent_name},
+                "userMessage": ev.get("body", ""),
+                "contactChannelRef": {"name": channel_name},
+                "threadID": ev.get("threadId", ""),
+            },
+            labels={"acp.humanlayer.dev/v1beta3": "true"},
+        )
+        store.create(task)
+        return JSONResponse({"taskName": task_name, "channelName": channel_name}, status_code=201)
+
+    # -------------------------------------------------------------- approvals
+
+    @app.get("/v1/approvals")
+    def list_approvals(namespace: str = "default"):
+        return [
+            {
+                "id": a["metadata"]["name"],
+                "type": a.get("spec", {}).get("type", ""),
+                "fn": a.get("spec", {}).get("fn", ""),
+                "message": a.get("spec", {}).get("message", ""),
+                "resolved": bool(a.get("status", {}).get("respondedAt")),
+            }
+            for a in store.list(APPROVAL, namespace)
+        ]
+
+    @app.post("/v1/approvals/{call_id}")
+    async def resolve_approval(call_id: str, request: Request, namespace: str = "default"):
+        body = json.loads(await request.body())
+        obj = store.get(APPROVAL, call_id, namespace)
+        if obj is None:
+            return JSONResponse({"error": "approval not found"}, status_code=404)
+        if obj.get("spec", {}).get("type") == "human_contact":
+            respond_to_contact(store, call_id, body.get("response", ""), namespace)
+        else:
+            respond_to_approval(
+                store, call_id, bool(body.get("approved")), body.get("comment", ""), namespace
+            )
+        return {"id": call_id, "resolved": True}
+
+    # ---------------------------------------------------------------- metrics
+
+    @app.get("/metrics")
+    def metrics():
+        lines = []
+        for kind, count in store.stats().items():
+            lines.append(f'acp_resources{{kind="{kind}"}} {count}')
+        if manager is not None:
+            for kind, depth in manager.queue_depths().items():
+                lines.append(f'acp_workqueue_depth{{kind="{kind}"}} {depth}')
+        if engine is not None and hasattr(engine, "metrics"):
+            for k, v in engine.metrics().items():
+                lines.append(f"acp_engine_{k} {v}")
+        return Response("\n".join(lines) + "\n", media_type="text/plain")
+
+    return app

@@ -1,0 +1,416 @@
+"""Durable resource store with watches — the etcd/apiserver role.
+
+The reference persists every resource and status transition to etcd through
+the Kubernetes apiserver, and controllers react through informer watches
+(acp/cmd/main.go:208-320 wires the manager's caches).  This
+module is the MI355X-native equivalent: a process-local, thread-safe,
+WAL-durable object store with
+
+- optimistic concurrency via ``metadata.resourceVersion`` (update conflicts
+  surface as ``ConflictError``, mirroring apiserver 409s that the reference's
+  conflict-retry loops handle, e.g. agent/state_machine.go:162-204),
+- label-filtered list (the reference lists ToolCalls by
+  ``{task, toolcallrequest}`` labels, task/state_machine.go:291-341),
+- watch streams (ADDED/MODIFIED/DELETED) delivered to subscriber queues —
+  these drive the controllers' workqueues, replacing the reference's
+  poll-with-RequeueAfter joins with watch-triggered reconciles,
+- an append-only JSONL WAL + snapshot for crash recovery (etcd's durability
+  role): every write is appended; ``load()`` replays.
+
+Status is a subresource: ``update_status`` only replaces ``status`` (like
+``client.Status().Update``), ``update`` only replaces spec/metadata.
+"""
+from __future__ import annotations
+
+import copy
+import dataclasses
+import json
+import os
+import queue
+import threading
+import time
+from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple
+
+from ..api.types import API_VERSION, EVENT, LEASE, new_object_meta, now_iso
+
+
+class NotFoundError(KeyError):
+    pass
+
+
+class AlreadyExistsError(ValueError):
+    pass
+
+
+class ConflictError(RuntimeError):
+    """resourceVersion mismatch — caller should re-get and retry."""
+
+
+@dataclasses.dataclass
+class WatchEvent:
+    type: str  # ADDED | MODIFIED | DELETED
+    kind: str
+    obj: Dict[str, Any]
+
+
+class _Watch:
+    __slots__ = ("kinds", "q", "alive")
+
+    def __init__(self, kinds: Optional[Iterable[str]]):
+        self.kinds = set(kinds) if kinds else None
+        self.q: "queue.Queue[WatchEvent]" = queue.Queue()
+        self.alive = True
+
+
+class ResourceStore:
+    """Thread-safe namespaced object store with watches and a WAL.
+
+    ``wal_path=None`` keeps everything in memory (unit tests, benchmarks that
+    measure the non-durable floor); with a path, every mutation is appended as
+    one JSON line and fsync'd per ``fsync`` policy ("always" | "interval" |
+    "never").
+    """
+
+    def __init__(
+        self,
+        wal_path: Optional[str] = None,
+        fsync: str = "interval",
+        fsync_interval_s: float = 0.05,
+    ):
+        self._lock = threading.RLock()
+        # kind -> namespace -> name -> obj
+        self._data: Dict[str, Dict[str, Dict[str, Dict[str, Any]]]] = {}
+        self._rv = 0
+        self._watches: List[_Watch] = []
+        self._wal_path = wal_path
+        self._wal_file = None
+        self._fsync = fsync
+        self._fsync_interval = fsync_interval_s
+        self._last_fsync = 0.0
+        if wal_path:
+            os.makedirs(os.path.dirname(os.path.abspath(wal_path)), exist_ok=True)
+            if os.path.exists(wal_path):
+                self._replay(wal_path)
+            self._wal_file = open(wal_path, "a", encoding="utf-8")
+
+    # ------------------------------------------------------------------ WAL
+
+    def _replay(self, path: str) -> None:
+        with open(path, "r", encoding="utf-8") as f:
+            for line in f:
+                line = line.strip()
+                if not line:
+                    continue
+                try:
+                    rec = json.loads(line)
+                except json.JSONDecodeError:
+                    continue  # torn tail write after a crash
+                op, obj = rec.get("op"), rec.get("obj")
+                if not obj:
+                    continue
+                kind = obj.get("kind")
+                m = obj.get("metadata", {})
+                ns, name = m.get("namespace", "default"), m.get("name")
+                if not kind or not name:
+                    continue
+                bucket = self._data.setdefault(kind, {}).setdefault(ns, {})
+                if op == "delete":
+                    bucket.pop(name, None)
+                else:
+                    bucket[name] = obj
+                self._rv = max(self._rv, int(m.get("resourceVersion", 0)))
+
+    def _append_wal(self, op: str, obj: Dict[str, Any]) -> None:
+        if self._wal_file is None:
+            return
+        self._wal_file.write(json.dumps({"op": op, "obj": obj}, separators=(",", ":")) + "\n")
+        self._wal_file.flush()
+        if self._fsync == "always":
+            os.fsync(self._wal_file.fileno())
+        elif self._fsync == "interval":
+            now = time.monotonic()
+            if now - self._last_fsync >= self._fsync_interval:
+                os.fsync(self._wal_file.fileno())
+                self._last_fsync = now
+
+    def compact(self) -> None:
+        """Rewrite the WAL as one snapshot line per live object."""
+        if not self._wal_path:
+            return
+        with self._lock:
+            tmp = self._wal_path + ".tmp"
+            with open(tmp, "w", encoding="utf-8") as f:
+                for kind, nss in self._data.items():
+                    for ns, objs in nss.items():
+                        for obj in objs.values():
+                            f.write(
+                                json.dumps({"op": "put", "obj": obj}, separators=(",", ":"))
+                                + "\n"
+                            )
+                f.flush()
+                os.fsync(f.fileno())
+            if self._wal_file:
+                self._wal_file.close()
+            os.replace(tmp, self._wal_path)
+            self._wal_file = open(self._wal_path, "a", encoding="utf-8")
+
+    def close(self) -> None:
+        with self._lock:
+            if self._wal_file:
+                self._wal_file.close()
+                self._wal_file = None
+
+    # ------------------------------------------------------------ primitives
+
+    def _notify(self, ev: WatchEvent) -> None:
+        for w in self._watches:
+            if w.alive and (w.kinds is None or ev.kind in w.kinds):
+                w.q.put(ev)
+
+    def create(self, obj: Dict[str, Any]) -> Dict[str, Any]:
+        obj = copy.deepcopy(obj)
+        kind = obj["kind"]
+        m = obj.setdefault("metadata", {})
+        ns = m.setdefault("namespace", "default")
+        name = m["name"]
+        if "uid" not in m or "creationTimestamp" not in m:
+            base = new_object_meta(name, ns)
+            m.setdefault("uid", base["uid"])
+            m.setdefault("creationTimestamp", base["creationTimestamp"])
+        obj.setdefault("apiVersion", API_VERSION)
+        obj.setdefault("spec", {})
+        obj.setdefault("status", {})
+        with self._lock:
+            bucket = self._data.setdefault(kind, {}).setdefault(ns, {})
+            if name in bucket:
+                raise AlreadyExistsError(f"{kind} {ns}/{name} already exists")
+            self._rv += 1
+            m["resourceVersion"] = self._rv
+            m["generation"] = 1
+            bucket[name] = obj
+            self._append_wal("put", obj)
+            out = copy.deepcopy(obj)
+            self._notify(WatchEvent("ADDED", kind, out))
+        return out
+
+    def get(self, kind: str, name: str, namespace: str = "default") -> Optional[Dict[str, Any]]:
+        with self._lock:
+            obj = self._data.get(kind, {}).get(namespace, {}).get(name)
+            return copy.deepcopy(obj) if obj is not None else None
+
+    def list(
+        self,
+        kind: str,
+        namespace: Optional[str] = "default",
+        label_selector: Optional[Dict[str, str]] = None,
+    ) -> List[Dict[str, Any]]:
+        with self._lock:
+            nss = self._data.get(kind, {})
+            spaces = [namespace] if namespace is not None else list(nss.keys())
+            out = []
+            for ns in spaces:
+                for obj in nss.get(ns, {}).values():
+                    if label_selector:
+                        labels = obj.get("metadata", {}).get("labels", {}) or {}
+                        if any(labels.get(k) != v for k, v in label_selector.items()):
+                            continue
+                    out.append(copy.deepcopy(obj))
+            return out
+
+    def _check_rv(self, current: Dict[str, Any], incoming: Dict[str, Any]) -> None:
+        want = incoming.get("metadata", {}).get("resourceVersion")
+        have = current.get("metadata", {}).get("resourceVersion")
+        if want is not None and int(want) != int(have):
+            raise ConflictError(
+                f'{incoming.get("kind")} {incoming["metadata"].get("name")}: '
+                f"resourceVersion conflict (have {have}, got {want})"
+            )
+
+    def update(self, obj: Dict[str, Any]) -> Dict[str, Any]:
+        """Replace spec + metadata (labels etc.); status is untouched."""
+        return self._update(obj, which="main")
+
+    def update_status(self, obj: Dict[str, Any]) -> Dict[str, Any]:
+        """Replace only status — the Status().Update subresource semantics."""
+        return self._update(obj, which="status")
+
+    def _update(self, obj: Dict[str, Any], which: str) -> Dict[str, Any]:
+        kind = obj["kind"]
+        m = obj.get("metadata", {})
+        ns, name = m.get("namespace", "default"), m["name"]
+        with self._lock:
+            bucket = self._data.get(kind, {}).get(ns, {})
+            cur = bucket.get(name)
+            if cur is None:
+                raise NotFoundError(f"{kind} {ns}/{name} not found")
+            self._check_rv(cur, obj)
+            self._rv += 1
+            if which == "status":
+                cur["status"] = copy.deepcopy(obj.get("status", {}))
+            else:
+                cur["spec"] = copy.deepcopy(obj.get("spec", {}))
+                newm = copy.deepcopy(obj.get("metadata", {}))
+                keep = cur["metadata"]
+                for k in ("labels", "annotations", "ownerReferences", "deletionTimestamp"):
+                    if k in newm:
+                        keep[k] = newm[k]
+                    elif k in keep and k not in newm:
+                        keep.pop(k, None)
+                keep["generation"] = int(keep.get("generation", 1)) + 1
+            cur["metadata"]["resourceVersion"] = self._rv
+            self._append_wal("put", cur)
+            out = copy.deepcopy(cur)
+            self._notify(WatchEvent("MODIFIED", kind, out))
+        return out
+
+    def delete(self, kind: str, name: str, namespace: str = "default") -> bool:
+        with self._lock:
+            bucket = self._data.get(kind, {}).get(namespace, {})
+            obj = bucket.pop(name, None)
+            if obj is None:
+                return False
+            self._append_wal("delete", obj)
+            self._notify(WatchEvent("DELETED", kind, copy.deepcopy(obj)))
+            # cascade: delete objects owned by this uid (k8s GC role)
+            uid = obj.get("metadata", {}).get("uid")
+            if uid:
+                for k2, nss in list(self._data.items()):
+                    for ns2, objs in list(nss.items()):
+                        for n2, o2 in list(objs.items()):
+                            for ref in o2.get("metadata", {}).get("ownerReferences", []) or []:
+                                if ref.get("uid") == uid:
+                                    self.delete(k2, n2, ns2)
+                                    break
+        return True
+
+    # --------------------------------------------------------------- watches
+
+    def watch(self, kinds: Optional[Iterable[str]] = None) -> "queue.Queue[WatchEvent]":
+        w = _Watch(kinds)
+        with self._lock:
+            self._watches.append(w)
+        return w.q
+
+    def stop_watch(self, q: "queue.Queue[WatchEvent]") -> None:
+        with self._lock:
+            for w in self._watches:
+                if w.q is q:
+                    w.alive = False
+            self._watches = [w for w in self._watches if w.alive]
+
+    # ---------------------------------------------------------------- events
+
+    def record_event(
+        self,
+        involved: Dict[str, Any],
+        event_type: str,
+        reason: str,
+        message: str,
+    ) -> None:
+        """Kubernetes-Events equivalent — the user-facing execution history
+        (the reference has ~40 recorder.Event call sites, e.g.
+        task/state_machine.go:224,628,731)."""
+        m = involved.get("metadata", {})
+        name = f'{m.get("name", "obj")}.{self._rv + 1}'
+        ev = {
+            "apiVersion": "v1",
+            "kind": EVENT,
+            "metadata": new_object_meta(name, m.get("namespace", "default")),
+            "spec": {},
+            "status": {},
+            "involvedObject": {
+                "kind": involved.get("kind"),
+                "name": m.get("name"),
+                "namespace": m.get("namespace", "default"),
+                "uid": m.get("uid", ""),
+            },
+            "type": event_type,
+            "reason": reason,
+            "message": message,
+            "lastTimestamp": now_iso(),
+        }
+        with self._lock:
+            bucket = self._data.setdefault(EVENT, {}).setdefault(
+                m.get("namespace", "default"), {}
+            )
+            self._rv += 1
+            ev["metadata"]["resourceVersion"] = self._rv
+            bucket[ev["metadata"]["name"]] = ev
+            self._append_wal("put", ev)
+            self._notify(WatchEvent("ADDED", EVENT, copy.deepcopy(ev)))
+
+    def events_for(self, involved_name: str, namespace: str = "default") -> List[Dict[str, Any]]:
+        with self._lock:
+            out = [
+                copy.deepcopy(e)
+                for e in self._data.get(EVENT, {}).get(namespace, {}).values()
+                if e.get("involvedObject", {}).get("name") == involved_name
+            ]
+            out.sort(key=lambda e: int(e["metadata"].get("resourceVersion", 0)))
+            return out
+
+    # ---------------------------------------------------------------- leases
+
+    def acquire_lease(
+        self,
+        name: str,
+        holder: str,
+        duration_s: float,
+        namespace: str = "default",
+    ) -> bool:
+        """coordination.k8s.io Lease semantics for task locking
+        (task/state_machine.go:1069-1145): acquire if absent, expired, or
+        held by the same holder (renew).  Returns False when held by another
+        live holder."""
+        now = time.time()
+        with self._lock:
+            bucket = self._data.setdefault(LEASE, {}).setdefault(namespace, {})
+            cur = bucket.get(name)
+            if cur is not None:
+                spec = cur.get("spec", {})
+                expired = now - float(spec.get("renewTime", 0)) > float(
+                    spec.get("leaseDurationSeconds", duration_s)
+                )
+                if spec.get("holderIdentity") != holder and not expired:
+                    return False
+            self._rv += 1
+            lease = {
+                "apiVersion": "coordination.k8s.io/v1",
+                "kind": LEASE,
+                "metadata": {
+                    "name": name,
+                    "namespace": namespace,
+                    "resourceVersion": self._rv,
+                },
+                "spec": {
+                    "holderIdentity": holder,
+                    "leaseDurationSeconds": duration_s,
+                    "renewTime": now,
+                },
+                "status": {},
+            }
+            bucket[name] = lease
+            self._append_wal("put", lease)
+            return True
+
+    def release_lease(self, name: str, holder: str, namespace: str = "default") -> None:
+        with self._lock:
+            bucket = self._data.get(LEASE, {}).get(namespace, {})
+            cur = bucket.get(name)
+            if cur is not None and cur.get("spec", {}).get("holderIdentity") == holder:
+                bucket.pop(name, None)
+                self._append_wal("delete", cur)
+
+    # ------------------------------------------------------------- utilities
+
+    def ensure_namespace(self, namespace: str) -> None:
+        # namespaces are implicit buckets; kept for API parity with createTask's
+        # ensureNamespaceExists (server.go)
+        return
+
+    def stats(self) -> Dict[str, int]:
+        with self._lock:
+            return {
+                kind: sum(len(objs) for objs in nss.values())
+                for kind, nss in self._data.items()
+            }

@@ -1,0 +1,137 @@
+"""Store tier: CRUD, optimistic concurrency, watches, WAL durability, leases.
+
+Mirrors the role of envtest in the reference test strategy:
+the store must behave like etcd+apiserver for everything the controllers
+rely on.
+"""
+import os
+
+import pytest
+
+from agentcontrolplane_amd.api.types import TASK, make_resource
+from agentcontrolplane_amd.store import (
+    AlreadyExistsError,
+    ConflictError,
+    ResourceStore,
+)
+
+
+def test_create_get_list(store):
+    obj = store.create(make_resource(TASK, "t1", spec={"agentRef": {"name": "a"}}))
+    assert obj["metadata"]["resourceVersion"] == 1
+    assert store.get(TASK, "t1")["spec"]["agentRef"]["name"] == "a"
+    assert len(store.list(TASK)) == 1
+    with pytest.raises(AlreadyExistsError):
+        store.create(make_resource(TASK, "t1"))
+
+
+def test_update_status_subresource(store):
+    obj = store.create(make_resource(TASK, "t1", spec={"x": 1}))
+    obj["status"]["phase"] = "Initializing"
+    store.update_status(obj)
+    got = store.get(TASK, "t1")
+    assert got["status"]["phase"] == "Initializing"
+    # spec update does not clobber status
+    got["spec"]["x"] = 2
+    store.update(got)
+    got2 = store.get(TASK, "t1")
+    assert got2["spec"]["x"] == 2 and got2["status"]["phase"] == "Initializing"
+
+
+def test_conflict_on_stale_rv(store):
+    obj = store.create(make_resource(TASK, "t1"))
+    first = store.get(TASK, "t1")
+    second = store.get(TASK, "t1")
+    first["status"]["phase"] = "A"
+    store.update_status(first)
+    second["status"]["phase"] = "B"
+    with pytest.raises(ConflictError):
+        store.update_status(second)
+
+
+def test_label_selector(store):
+    store.create(make_resource(TASK, "t1", labels={"team": "a"}))
+    store.create(make_resource(TASK, "t2", labels={"team": "b"}))
+    assert [t["metadata"]["name"] for t in store.list(TASK, label_selector={"team": "a"})] == ["t1"]
+
+
+def test_watch_events(store):
+    q = store.watch([TASK])
+    store.create(make_resource(TASK, "t1"))
+    ev = q.get(timeout=1)
+    assert ev.type == "ADDED" and ev.obj["metadata"]["name"] == "t1"
+    obj = store.get(TASK, "t1")
+    obj["status"]["phase"] = "X"
+    store.update_status(obj)
+    assert q.get(timeout=1).type == "MODIFIED"
+    store.delete(TASK, "t1")
+    assert q.get(timeout=1).type == "DELETED"
+
+
+def test_owner_cascade_delete(store):
+    from agentcontrolplane_amd.api.types import TOOL_CALL, owner_ref
+
+    parent = store.create(make_resource(TASK, "t1"))
+    child = make_resource(TOOL_CALL, "tc1")
+    child["metadata"]["ownerReferences"] = [owner_ref(parent)]
+    store.create(child)
+    store.delete(TASK, "t1")
+    assert store.get(TOOL_CALL, "tc1") is None
+
+
+def test_wal_persistence(tmp_path):
+    wal = str(tmp_path / "acp.wal")
+    s1 = ResourceStore(wal_path=wal, fsync="always")
+    s1.create(make_resource(TASK, "t1", spec={"k": "v"}))
+    obj = s1.get(TASK, "t1")
+    obj["status"]["phase"] = "ReadyForLLM"
+    obj["status"]["contextWindow"] = [{"role": "user", "content": "hi"}]
+    s1.update_status(obj)
+    s1.close()
+    # crash-resume: replay the WAL
+    s2 = ResourceStore(wal_path=wal)
+    got = s2.get(TASK, "t1")
+    assert got["status"]["phase"] == "ReadyForLLM"
+    assert got["status"]["contextWindow"][0]["content"] == "hi"
+    s2.close()
+
+
+def test_wal_compaction(tmp_path):
+    wal = str(tmp_path / "acp.wal")
+    s = ResourceStore(wal_path=wal)
+    obj = s.create(make_resource(TASK, "t1"))
+    for i in range(50):
+        obj = s.get(TASK, "t1")
+        obj["status"]["phase"] = f"p{i}"
+        s.update_status(obj)
+    before = os.path.getsize(wal)
+    s.compact()
+    after = os.path.getsize(wal)
+    assert after < before
+    s.close()
+    s2 = ResourceStore(wal_path=wal)
+    assert s2.get(TASK, "t1")["status"]["phase"] == "p49"
+    s2.close()
+
+
+def test_lease_semantics(store):
+    assert store.acquire_lease("task-llm-t1", "pod-a", duration_s=30)
+    assert not store.acquire_lease("task-llm-t1", "pod-b", duration_s=30)
+    assert store.acquire_lease("task-llm-t1", "pod-a", duration_s=30)  # renew
+    store.release_lease("task-llm-t1", "pod-a")
+    assert store.acquire_lease("task-llm-t1", "pod-b", duration_s=30)
+
+
+def test_lease_expiry(store):
+    assert store.acquire_lease("l", "pod-a", duration_s=0.01)
+    import time
+
+    time.sleep(0.05)
+    assert store.acquire_lease("l", "pod-b", duration_s=30)
+
+
+def test_events_recorded(store):
+    obj = store.create(make_resource(TASK, "t1"))
+    store.record_event(obj, "Normal", "ValidationSucceeded", "ok")
+    evs = store.events_for("t1")
+    assert len(evs) == 1 and evs[0]["reason"] == "ValidationSucceeded"
