@@ -331,7 +331,7 @@ def build_app(store, manager=None, engine=None) -> FastAPI:
 
     @app.get("/metrics")
     def metrics():
-        lines = []
+        lines = ["acp_up 1"]
         for kind, count in store.stats().items():
             lines.append(f'acp_resources{{kind="{kind}"}} {count}')
         if manager is not None:

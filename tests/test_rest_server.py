@@ -168,4 +168,4 @@ def test_approvals_endpoint(client, cp):
 def test_metrics_endpoint(client, cp):
     r = client.get("/metrics")
     assert r.status_code == 200
-    assert "acp_resources" in r.text
+    assert "acp_up 1" in r.text
