@@ -1,0 +1,381 @@
+"""Grammar-constrained decoding for tool-call JSON.
+
+The reference gets syntactically valid tool calls for free from the remote
+provider; an in-process engine must *enforce* them.  This module implements
+a byte-level pushdown automaton that constrains sampling to
+
+    {"name": "<one of the offered tool names>", "arguments": <JSON object>}
+
+per tool call.  At every decode step the automaton exposes the set of legal
+next bytes; the sampler masks the logits to that set (plus EOT when the
+automaton is in an accepting state).  With the byte-level tokenizer one
+token = one byte, so the automaton advances one state per sampled token.
+
+States use an explicit stack so arbitrarily nested argument objects are
+legal up to ``max_depth``; ``max_len`` bounds runaway generations.
+"""
+from __future__ import annotations
+
+import json
+from typing import Dict, Iterable, List, Optional, Set, Tuple
+
+from .tokenizer import EOT
+
+WS = b" \t\n\r"
+DIGITS = b"0123456789"
+HEX = b"0123456789abcdefABCDEF"
+# printable ASCII minus '"' and '\': generation stays single-byte-valid UTF-8
+# (multi-byte sequences would need a UTF-8 sub-automaton to stay decodable)
+STRING_SAFE = bytes(b for b in range(0x20, 0x7F) if b not in (0x22, 0x5C))
+
+
+def _bs(*parts: Iterable[int]) -> Set[int]:
+    out: Set[int] = set()
+    for p in parts:
+        out.update(p)
+    return out
+
+
+class JsonValueMachine:
+    """Byte-level PDA accepting one JSON value.  ``allowed()`` returns legal
+    next bytes; ``done`` is True in an accepting state (value complete)."""
+
+    def __init__(self, max_depth: int = 8, max_len: int = 4096, root_object: bool = False):
+        self.stack: List[str] = []
+        self.state = "value"          # expecting a value
+        self.max_depth = max_depth
+        self.max_len = max_len
+        self.count = 0
+        self.done = False
+        self.root_object = root_object
+
+    # states:
+    #   value        expecting start of a value
+    #   string       inside a string
+    #   str_escape   after backslash in string
+    #   str_u{n}     expecting n more hex digits
+    #   num_*        number sub-states
+    #   lit:<rest>   finishing a literal (true/false/null)
+    #   obj_key      expecting '"' of a key (or '}' right after '{')
+    #   obj_colon    expecting ':'
+    #   obj_next     expecting ',' or '}'
+    #   arr_next     expecting ',' or ']'
+    # stack frames: 'O' (in object, after value), 'A' (in array, after value)
+
+    def allowed(self) -> Set[int]:
+        if self.count >= self.max_len:
+            return self._closing_allowed()
+        s = self.state
+        deep = len(self.stack) >= self.max_depth
+        if s == "value":
+            if self.root_object and not self.stack:
+                return _bs(b"{")  # the root value must be an object
+            opts = _bs(WS, b'"', DIGITS, b"-", b"tfn")
+            if not deep:
+                opts |= _bs(b"{[")
+            return opts
+        if s == "string":
+            return _bs(STRING_SAFE, b'"', b"\\")
+        if s == "str_escape":
+            return _bs(b'"\\/bfnrtu')
+        if s.startswith("str_u"):
+            return _bs(HEX)
+        if s == "num_int":
+            # after '-' or first digits
+            return _bs(DIGITS, b".eE") | self._terminators()
+        if s == "num_start":
+            return _bs(DIGITS)
+        if s == "num_frac_first":
+            return _bs(DIGITS)
+        if s == "num_frac":
+            return _bs(DIGITS, b"eE") | self._terminators()
+        if s == "num_exp_sign":
+            return _bs(DIGITS, b"+-")
+        if s == "num_exp_first":
+            return _bs(DIGITS)
+        if s == "num_exp":
+            return _bs(DIGITS) | self._terminators()
+        if s.startswith("lit:"):
+            return {s[4:].encode()[0]}
+        if s == "obj_key_first":
+            return _bs(WS, b'"', b"}")
+        if s == "obj_key":
+            return _bs(WS, b'"')
+        if s == "obj_key_str":
+            return _bs(STRING_SAFE, b'"', b"\\")
+        if s == "obj_key_escape":
+            return _bs(b'"\\/bfnrtu')
+        if s == "obj_colon":
+            return _bs(WS, b":")
+        if s == "obj_next":
+            return _bs(WS, b",}")
+        if s == "arr_next":
+            return _bs(WS, b",]")
+        if s == "arr_first":
+            opts = _bs(WS, b'"', DIGITS, b"-", b"tfn", b"]")
+            if len(self.stack) < self.max_depth:
+                opts |= _bs(b"{[")
+            return opts
+        if s == "done":
+            return set()
+        raise AssertionError(f"bad state {s}")
+
+    def _closing_allowed(self) -> Set[int]:
+        """Over the length budget: only bytes on the shortest path to an
+        accepting state, so generation always terminates in valid JSON."""
+        s = self.state
+        if s in ("string", "obj_key_str"):
+            return {0x22}  # '"'
+        if s in ("str_escape", "obj_key_escape"):
+            return {ord("n")}
+        if s.startswith("str_u"):
+            return {ord("0")}
+        if s in ("num_start", "num_frac_first", "num_exp_first", "num_exp_sign"):
+            return {ord("0")}
+        if s in ("num_int", "num_frac", "num_exp"):
+            t = self._terminators() - _bs(WS, b",")
+            return t or {ord("0")}
+        if s.startswith("lit:"):
+            return {s[4:].encode()[0]}
+        if s in ("value",):
+            return {ord("0")}
+        if s in ("obj_key_first", "obj_next"):
+            return {ord("}")}
+        if s == "obj_key":
+            return {0x22}
+        if s == "obj_colon":
+            return {ord(":")}
+        if s in ("arr_first", "arr_next"):
+            return {ord("]")}
+        return set()
+
+    def _terminators(self) -> Set[int]:
+        """Bytes that may legally follow a number given the stack."""
+        if not self.stack:
+            return set()
+        if self.stack[-1] == "O":
+            return _bs(WS, b",}")
+        return _bs(WS, b",]")
+
+    def advance(self, b: int) -> None:
+        self.count += 1
+        s = self.state
+        c = bytes([b])
+        if s == "value":
+            if c in (b" ", b"\t", b"\n", b"\r"):
+                self.count -= 0
+                return
+            if c == b'"':
+                self.state = "string"
+            elif c == b"{":
+                self.stack.append("O")
+                self.state = "obj_key_first"
+            elif c == b"[":
+                self.stack.append("A")
+                self.state = "arr_value_or_end"
+                # treat immediately: allow value or ']' — model as value with ']' option
+                self.state = "arr_first"
+            elif c == b"-":
+                self.state = "num_start"
+            elif b in DIGITS:
+                self.state = "num_int"
+            elif c == b"t":
+                self.state = "lit:rue"
+            elif c == b"f":
+                self.state = "lit:alse"
+            elif c == b"n":
+                self.state = "lit:ull"
+            else:
+                raise ValueError(f"illegal byte {c!r} in state {s}")
+            return
+        if s == "arr_first":
+            if c in (b" ", b"\t", b"\n", b"\r"):
+                return
+            if c == b"]":
+                self.stack.pop()
+                self._value_done()
+                return
+            # else it's the start of a value — re-dispatch through 'value'
+            self.state = "value"
+            self.count -= 1
+            return self.advance(b)
+        if s == "string":
+            if c == b'"':
+                self._value_done()
+            elif c == b"\\":
+                self.state = "str_escape"
+            return
+        if s == "str_escape":
+            if c == b"u":
+                self.state = "str_u4"
+            else:
+                self.state = "string"
+            return
+        if s.startswith("str_u"):
+            n = int(s[5:])
+            self.state = "string" if n == 1 else f"str_u{n - 1}"
+            return
+        if s in ("num_start", "num_frac_first", "num_exp_first", "num_exp_sign"):
+            if s == "num_start":
+                self.state = "num_int"
+            elif s == "num_frac_first":
+                self.state = "num_frac"
+            elif s == "num_exp_sign" and c in b"+-":
+                self.state = "num_exp_first"
+            else:
+                self.state = "num_exp"
+            return
+        if s in ("num_int", "num_frac", "num_exp"):
+            if b in DIGITS:
+                return
+            if c == b"." and s == "num_int":
+                self.state = "num_frac_first"
+                return
+            if c in b"eE" and s in ("num_int", "num_frac"):
+                self.state = "num_exp_sign"
+                return
+            # terminator: the number is done; re-dispatch
+            self._value_done()
+            self.count -= 1
+            return self.advance(b)
+        if s.startswith("lit:"):
+            rest = s[4:]
+            if c != rest[:1].encode():
+                raise ValueError(f"illegal literal byte {c!r}")
+            self.state = f"lit:{rest[1:]}" if len(rest) > 1 else "__lit_done__"
+            if self.state == "__lit_done__":
+                self._value_done()
+            return
+        if s in ("obj_key", "obj_key_first"):
+            if c in (b" ", b"\t", b"\n", b"\r"):
+                return
+            if c == b'"':
+                self.state = "obj_key_str"
+            elif c == b"}" and s == "obj_key_first":
+                self.stack.pop()
+                self._value_done()
+            return
+        if s == "obj_key_str":
+            if c == b'"':
+                self.state = "obj_colon"
+            elif c == b"\\":
+                self.state = "obj_key_escape"
+            return
+        if s == "obj_key_escape":
+            self.state = "obj_key_str"
+            return
+        if s == "obj_colon":
+            if c == b":":
+                self.state = "value"
+            return
+        if s == "obj_next":
+            if c == b",":
+                self.state = "obj_key"
+            elif c == b"}":
+                self.stack.pop()
+                self._value_done()
+            return
+        if s == "arr_next":
+            if c == b",":
+                self.state = "value"
+            elif c == b"]":
+                self.stack.pop()
+                self._value_done()
+            return
+        raise ValueError(f"illegal byte {c!r} in state {s}")
+
+    def _value_done(self) -> None:
+        if not self.stack:
+            self.state = "done"
+            self.done = True
+        elif self.stack[-1] == "O":
+            self.state = "obj_next"
+        else:
+            self.state = "arr_next"
+
+
+class ToolCallGrammar:
+    """Constrains one tool call:
+    ``{"name": "<tool>", "arguments": {…}}`` then EOT.
+
+    ``allowed_tokens()`` returns legal token ids (bytes plus EOT in the
+    accepting state); ``advance(token)`` consumes the sampled token.
+    ``parse()`` returns the finished (name, arguments_json) pair.
+    """
+
+    PRE = b'{"name": "'
+    MID = b'", "arguments": '
+
+    def __init__(self, tool_names: List[str], max_args_len: int = 2048):
+        if not tool_names:
+            raise ValueError("no tools to constrain to")
+        self.names = sorted(set(tool_names))
+        self.buf = bytearray()
+        self.phase = "pre"       # pre → name → mid → args → done
+        self.pos = 0
+        self.name_prefix = b""
+        self.args = JsonValueMachine(max_len=max_args_len, root_object=True)
+        self.finished = False
+
+    def allowed_tokens(self) -> Set[int]:
+        if self.phase == "pre":
+            return {self.PRE[self.pos]}
+        if self.phase == "name":
+            nexts: Set[int] = set()
+            for n in self.names:
+                nb = n.encode()
+                if nb.startswith(self.name_prefix):
+                    if len(nb) > len(self.name_prefix):
+                        nexts.add(nb[len(self.name_prefix)])
+                    else:
+                        nexts.add(ord('"'))  # exact match may close
+            return nexts
+        if self.phase == "mid":
+            return {self.MID[self.pos]}
+        if self.phase == "args":
+            allowed = set(self.args.allowed())
+            if self.args.done:
+                allowed.add(ord("}"))
+            return allowed
+        if self.phase == "close":
+            return {ord("}")}
+        return {EOT}
+
+    def advance(self, token: int) -> None:
+        if token == EOT and self.phase == "done":
+            self.finished = True
+            return
+        b = token
+        self.buf.append(b)
+        if self.phase == "pre":
+            self.pos += 1
+            if self.pos == len(self.PRE):
+                self.phase = "name"
+            return
+        if self.phase == "name":
+            if b == ord('"') and self.name_prefix.decode() in self.names:
+                self.phase = "mid"
+                self.pos = 1  # the '"' consumed is MID[0]... MID starts with '"'
+                return
+            self.name_prefix += bytes([b])
+            return
+        if self.phase == "mid":
+            self.pos += 1
+            if self.pos == len(self.MID):
+                self.phase = "args"
+            return
+        if self.phase == "args":
+            if self.args.done and b == ord("}"):
+                self.phase = "done"
+                return
+            self.args.advance(b)
+            return
+        raise ValueError(f"unexpected token {token} in phase {self.phase}")
+
+    @property
+    def accepting(self) -> bool:
+        return self.phase == "done"
+
+    def parse(self) -> Tuple[str, str]:
+        obj = json.loads(bytes(self.buf).decode("utf-8"))
+        return obj["name"], json.dumps(obj["arguments"])

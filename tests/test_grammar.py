@@ -1,0 +1,88 @@
+"""Constrained-decoding grammar: every random walk through the automaton must
+produce parseable tool-call JSON naming an offered tool."""
+import json
+import random
+
+import pytest
+
+from agentcontrolplane_amd.engine.grammar import JsonValueMachine, ToolCallGrammar
+from agentcontrolplane_amd.engine.tokenizer import EOT
+
+
+def drive(grammar, rng, max_steps=3000):
+    steps = 0
+    while not grammar.finished and steps < max_steps:
+        allowed = grammar.allowed_tokens()
+        assert allowed, f"dead end in phase {grammar.phase}"
+        tok = rng.choice(sorted(allowed))
+        grammar.advance(tok)
+        steps += 1
+        if grammar.phase == "done":
+            grammar.advance(EOT)
+    assert grammar.finished
+    return bytes(grammar.buf).decode()
+
+
+@pytest.mark.parametrize("seed", range(20))
+def test_random_walks_parse(seed):
+    rng = random.Random(seed)
+    g = ToolCallGrammar(["calc__add", "calc__echo", "search__web"], max_args_len=200)
+    text = drive(g, rng)
+    obj = json.loads(text)
+    assert obj["name"] in ("calc__add", "calc__echo", "search__web")
+    assert isinstance(obj["arguments"], dict)
+    name, args = g.parse()
+    assert name == obj["name"]
+    json.loads(args)
+
+
+def test_forced_sequence():
+    g = ToolCallGrammar(["add"])
+    target = b'{"name": "add", "arguments": {"a": 1, "b": 2.5}}'
+    for b in target:
+        assert b in g.allowed_tokens(), f"byte {chr(b)!r} rejected, phase={g.phase}"
+        g.advance(b)
+    assert g.accepting
+    assert EOT in g.allowed_tokens()
+    g.advance(EOT)
+    name, args = g.parse()
+    assert name == "add"
+    assert json.loads(args) == {"a": 1, "b": 2.5}
+
+
+def test_shared_prefix_tool_names():
+    g = ToolCallGrammar(["calc__add", "calc__add_more"])
+    for b in b'{"name": "calc__add':
+        g.advance(b)
+    # both continuing and closing must be legal at the shared prefix
+    allowed = g.allowed_tokens()
+    assert ord('"') in allowed and ord("_") in allowed
+
+
+def test_illegal_name_rejected():
+    g = ToolCallGrammar(["add"])
+    for b in b'{"name": "':
+        g.advance(b)
+    assert ord("z") not in g.allowed_tokens()
+
+
+def test_json_value_machine_nested():
+    m = JsonValueMachine(root_object=True)
+    for b in b'{"x": [1, {"y": "z\\n"}, true, null, -2.5e3]}':
+        assert b in m.allowed(), f"rejected {chr(b)!r} in state {m.state}"
+        m.advance(b)
+    assert m.done
+
+
+def test_json_depth_limit():
+    m = JsonValueMachine(max_depth=2, root_object=True)
+    for b in b'{"a": {':
+        m.advance(b)
+    # at depth 2, '{' and '[' are no longer offered for values
+    allowed = m.allowed()
+    for b in b'"k':
+        pass
+    m.advance(ord('"'))
+    for b in b'k": ':
+        m.advance(b)
+    assert ord("{") not in m.allowed() and ord("[") not in m.allowed()
