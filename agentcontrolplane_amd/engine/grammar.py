@@ -81,8 +81,11 @@ class JsonValueMachine:
         if s.startswith("str_u"):
             return _bs(HEX)
         if s == "num_int":
-            # after '-' or first digits
+            # after first nonzero digits
             return _bs(DIGITS, b".eE") | self._terminators()
+        if s == "num_zero":
+            # leading zero: no further digits (JSON forbids 06)
+            return _bs(b".eE") | self._terminators()
         if s == "num_start":
             return _bs(DIGITS)
         if s == "num_frac_first":
@@ -104,7 +107,8 @@ class JsonValueMachine:
         if s == "obj_key_str":
             return _bs(STRING_SAFE, b'"', b"\\")
         if s == "obj_key_escape":
-            return _bs(b'"\\/bfnrtu')
+            # no \u in generated keys (obj_key_escape has no hex sub-states)
+            return _bs(b'"\\/bfnrt')
         if s == "obj_colon":
             return _bs(WS, b":")
         if s == "obj_next":
@@ -132,7 +136,7 @@ class JsonValueMachine:
             return {ord("0")}
         if s in ("num_start", "num_frac_first", "num_exp_first", "num_exp_sign"):
             return {ord("0")}
-        if s in ("num_int", "num_frac", "num_exp"):
+        if s in ("num_int", "num_zero", "num_frac", "num_exp"):
             t = self._terminators() - _bs(WS, b",")
             return t or {ord("0")}
         if s.startswith("lit:"):
@@ -177,6 +181,8 @@ class JsonValueMachine:
                 self.state = "arr_first"
             elif c == b"-":
                 self.state = "num_start"
+            elif c == b"0":
+                self.state = "num_zero"
             elif b in DIGITS:
                 self.state = "num_int"
             elif c == b"t":
@@ -217,7 +223,7 @@ class JsonValueMachine:
             return
         if s in ("num_start", "num_frac_first", "num_exp_first", "num_exp_sign"):
             if s == "num_start":
-                self.state = "num_int"
+                self.state = "num_zero" if c == b"0" else "num_int"
             elif s == "num_frac_first":
                 self.state = "num_frac"
             elif s == "num_exp_sign" and c in b"+-":
@@ -225,13 +231,13 @@ class JsonValueMachine:
             else:
                 self.state = "num_exp"
             return
-        if s in ("num_int", "num_frac", "num_exp"):
-            if b in DIGITS:
+        if s in ("num_int", "num_zero", "num_frac", "num_exp"):
+            if b in DIGITS and s != "num_zero":
                 return
-            if c == b"." and s == "num_int":
+            if c == b"." and s in ("num_int", "num_zero"):
                 self.state = "num_frac_first"
                 return
-            if c in b"eE" and s in ("num_int", "num_frac"):
+            if c in b"eE" and s in ("num_int", "num_zero", "num_frac"):
                 self.state = "num_exp_sign"
                 return
             # terminator: the number is done; re-dispatch
