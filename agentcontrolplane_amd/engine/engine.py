@@ -1,0 +1,274 @@
+"""The inference engine: model + paged KV + continuous batching loop.
+
+This replaces the reference's remote-LLM HTTP path (llmclient/
+langchaingo_client.go): ``chat()`` turns an ACP context window into token
+ids, submits an InferenceRequest, and blocks on its future while the engine
+thread continuously batches every in-flight request into shared GPU steps.
+
+Sampling is over the tokenizer's live vocabulary (bytes + specials); the
+full-vocab LM-head GEMM is still computed (that cost is real and stays in
+the measured path), the slice only bounds the sampling distribution to
+decodable ids.  Constrained (tool-call) sequences additionally mask to the
+grammar's legal next bytes each step.
+"""
+from __future__ import annotations
+
+import json
+import threading
+import time
+from typing import Any, Dict, List, Optional
+
+import torch
+
+from .. import ops
+from ..models import create_model
+from .config import EngineConfig
+from .grammar import ToolCallGrammar
+from .kv import make_block_manager
+from .request import ChatResult, InferenceRequest, SamplingParams
+from .scheduler import DECODE, Scheduler, Sequence
+from .tokenizer import EOT, N_SPECIAL, TOOL_CALL_START, ByteTokenizer
+
+
+class InferenceEngine:
+    def __init__(self, config: EngineConfig, model=None, start: bool = True):
+        self.cfg = config
+        self.mcfg = config.model_config()
+        self.device = torch.device(config.device)
+        self.tokenizer = ByteTokenizer(self.mcfg.vocab_size)
+        if model is not None:
+            self.model = model
+        else:
+            self.model = create_model(self.mcfg, config, config.device)
+            self.model.random_init(config.seed)
+        num_blocks = self._size_kv_pool()
+        self.model.allocate_kv_cache(num_blocks, config.kv_block_size)
+        self.bm = make_block_manager(
+            num_blocks, config.kv_block_size, prefer_native=True
+        )
+        self.scheduler = Scheduler(config, self.bm, self.device)
+        self._gen = torch.Generator(device=self.device.type)
+        self._gen.manual_seed(config.seed)
+        self._lock = threading.Lock()
+        self._work = threading.Condition(self._lock)
+        self._pending: List[InferenceRequest] = []
+        self._running = False
+        self._thread: Optional[threading.Thread] = None
+        # metrics
+        self._m = {
+            "steps": 0,
+            "prompt_tokens": 0,
+            "generated_tokens": 0,
+            "preemptions": 0,
+            "requests_completed": 0,
+            "busy_time_s": 0.0,
+        }
+        self._start_time = time.monotonic()
+        if start:
+            self.start()
+
+    # ------------------------------------------------------------- sizing
+
+    def _size_kv_pool(self) -> int:
+        if self.cfg.num_kv_blocks is not None:
+            return self.cfg.num_kv_blocks
+        if self.device.type != "cuda":
+            return 1024
+        free, total = torch.cuda.mem_get_info(self.device)
+        budget = int(total * self.cfg.gpu_memory_utilization) - (total - free)
+        block_bytes = self.model.kv_block_bytes(self.cfg.kv_block_size)
+        # keep headroom for activations: 4 GB or 5% of total
+        headroom = max(4 << 30, int(0.05 * total))
+        n = max(64, (budget - headroom) // block_bytes)
+        return int(n)
+
+    # ----------------------------------------------------------- lifecycle
+
+    def start(self) -> None:
+        if self._running:
+            return
+        self._running = True
+        self._thread = threading.Thread(target=self._loop, name="acp-engine", daemon=True)
+        self._thread.start()
+
+    def stop(self) -> None:
+        with self._work:
+            self._running = False
+            self._work.notify_all()
+        if self._thread is not None:
+            self._thread.join(timeout=10)
+            self._thread = None
+
+    # ------------------------------------------------------------- submit
+
+    def submit(self, request: InferenceRequest) -> InferenceRequest:
+        with self._work:
+            if len(self._pending) >= self.cfg.max_queue:
+                raise RuntimeError("engine queue full")
+            self._pending.append(request)
+            self._work.notify_all()
+        return request
+
+    def generate(
+        self,
+        prompt_ids: List[int],
+        sampling: Optional[SamplingParams] = None,
+        constrained: bool = False,
+        tools: Optional[List[Dict]] = None,
+        timeout: Optional[float] = None,
+    ) -> InferenceRequest:
+        req = InferenceRequest(prompt_ids, sampling or SamplingParams(), constrained, tools)
+        self.submit(req)
+        req.wait(timeout or self.cfg.request_timeout_s)
+        return req
+
+    def chat(
+        self,
+        messages: List[Dict[str, Any]],
+        tools: Optional[List[Dict[str, Any]]] = None,
+        sampling: Optional[SamplingParams] = None,
+    ) -> ChatResult:
+        """The LLMClient.send_request counterpart."""
+        t0 = time.monotonic()
+        sampling = sampling or SamplingParams()
+        tools = tools or []
+        prompt_ids = self.tokenizer.render_chat(messages, tools)
+        constrained = False
+        if tools:
+            choice = sampling.tool_choice
+            if choice == "auto":
+                # synthetic-workload policy (random-init weights cannot decide):
+                # call a tool once per conversation, then answer.  A trained
+                # checkpoint resolves "auto" from the TOOL_CALL_START logit.
+                has_tool_result = any(m.get("role") == "tool" for m in messages)
+                constrained = not has_tool_result
+            elif choice == "required":
+                constrained = True
+        req = InferenceRequest(prompt_ids, sampling, constrained, tools)
+        self.submit(req)
+        out_ids = req.wait(self.cfg.request_timeout_s)
+        result = ChatResult(
+            prompt_tokens=len(prompt_ids),
+            completion_tokens=len(out_ids),
+            finish_reason=req.finish_reason,
+            latency_s=time.monotonic() - t0,
+        )
+        if req.finish_reason == "tool_calls" and req.seq is not None and req.seq.grammar:
+            name, args = req.seq.grammar.parse()
+            result.tool_calls = [
+                {
+                    "id": f"call_{req.request_id:08d}",
+                    "type": "function",
+                    "function": {"name": name, "arguments": args},
+                }
+            ]
+        else:
+            result.text = self.tokenizer.decode(out_ids)
+        return result
+
+    # ------------------------------------------------------------ the loop
+
+    def _loop(self) -> None:
+        while True:
+            with self._work:
+                while self._running and not self._pending and not self.scheduler.has_work():
+                    self._work.wait(timeout=0.2)
+                if not self._running:
+                    return
+                pending, self._pending = self._pending, []
+            for req in pending:
+                self.scheduler.add_request(req)
+            try:
+                t0 = time.monotonic()
+                worked = self.step()
+                self._m["busy_time_s"] += time.monotonic() - t0
+            except Exception as e:  # noqa: BLE001 — fail all in-flight requests
+                import traceback
+
+                traceback.print_exc()
+                for s in list(self.scheduler.running) + list(self.scheduler.waiting):
+                    self.scheduler.abort(s, e)
+
+    def step(self) -> bool:
+        """One scheduler round + forward + sampling.  Returns True if any
+        work ran."""
+        out = self.scheduler.schedule()
+        self._m["preemptions"] += len(out.preempted)
+        if out.batch is None:
+            return False
+        batch = out.batch
+        logits = self.model.forward(batch)
+        self._sample_and_commit(batch, logits)
+        self._m["steps"] += 1
+        self._m["prompt_tokens"] += batch.num_prefill_tokens
+        return True
+
+    def _sample_and_commit(self, batch, logits: torch.Tensor) -> None:
+        seqs: List[Sequence] = []
+        for sid in batch.sample_seq_ids:
+            seqs.append(self.scheduler.seq_by_id(sid))
+        B = len(seqs)
+        if B == 0:
+            return
+        live = logits[:, :N_SPECIAL]  # sampling restricted to decodable ids
+        temps = torch.tensor(
+            [s.request.sampling.temperature for s in seqs], device=logits.device
+        )
+        top_ks = torch.tensor(
+            [s.request.sampling.top_k for s in seqs], device=logits.device, dtype=torch.long
+        )
+        top_ps = torch.tensor([s.request.sampling.top_p for s in seqs], device=logits.device)
+        mask = None
+        any_grammar = any(s.grammar is not None for s in seqs)
+        if any_grammar:
+            mask = torch.ones(B, N_SPECIAL, dtype=torch.bool)
+            for i, s in enumerate(seqs):
+                if s.grammar is not None:
+                    allowed = s.grammar.allowed_tokens()
+                    if s.grammar.accepting:
+                        allowed = set(allowed) | {EOT}
+                    row = torch.zeros(N_SPECIAL, dtype=torch.bool)
+                    row[list(allowed)] = True
+                    mask[i] = row
+            mask = mask.to(logits.device)
+        tokens = ops.softmax_sample(live, temps, top_ks, top_ps, self._gen, mask)
+        tokens_cpu = tokens.tolist()
+        for s, tok in zip(seqs, tokens_cpu):
+            self.scheduler.append_sampled(s, tok)
+            self._m["generated_tokens"] += 1
+            if s.request.first_token_time is None:
+                s.request.first_token_time = time.monotonic()
+            if s.grammar is not None:
+                if tok == EOT and s.grammar.accepting:
+                    s.grammar.advance(EOT)
+                    s.request.output_ids.pop()  # EOT is not part of the JSON
+                    s.output_ids.pop()
+                    self._m["requests_completed"] += 1
+                    self.scheduler.finish_seq(s, "tool_calls")
+                    continue
+                s.grammar.advance(tok)
+                if len(s.output_ids) >= s.request.sampling.max_tokens * 8:
+                    # runaway guard; grammar's closing mode should prevent this
+                    self._m["requests_completed"] += 1
+                    self.scheduler.finish_seq(s, "length")
+                continue
+            if tok == EOT:
+                s.request.output_ids.pop()
+                s.output_ids.pop()
+                self._m["requests_completed"] += 1
+                self.scheduler.finish_seq(s, "stop")
+            elif len(s.output_ids) >= s.request.sampling.max_tokens:
+                self._m["requests_completed"] += 1
+                self.scheduler.finish_seq(s, "length")
+
+    # ------------------------------------------------------------- metrics
+
+    def metrics(self) -> Dict[str, float]:
+        m = dict(self._m)
+        elapsed = time.monotonic() - self._start_time
+        m["uptime_s"] = elapsed
+        m["tokens_per_s"] = self._m["generated_tokens"] / max(1e-9, elapsed)
+        m["kv_occupancy"] = self.bm.used_blocks / max(1, self.bm.num_blocks)
+        m["running_seqs"] = len(self.scheduler.running)
+        m["waiting_seqs"] = len(self.scheduler.waiting)
+        return m

@@ -1,0 +1,111 @@
+"""Paged KV-cache block management.
+
+The engine's KV cache is paged vLLM-style: per layer a pool of fixed-size
+blocks (``block_size`` tokens each); every sequence holds an ordered block
+table.  The pool is sized for the GPU's free HBM (288 GB on MI355X) at
+engine start.
+
+The allocator itself is native C++ (csrc/block_manager.cpp, exposed through
+the ``_C`` extension) — free-list allocation, block-table bookkeeping and
+slot-mapping computation run outside the GIL for thousands of concurrent
+sequences.  ``PyBlockManager`` is the pure-python equivalent used on CPU and
+as the numerics oracle for the C++ one (tests assert identical behavior).
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+
+class OutOfBlocksError(RuntimeError):
+    pass
+
+
+class PyBlockManager:
+    """Free-list block allocator + per-sequence block tables."""
+
+    def __init__(self, num_blocks: int, block_size: int):
+        self.num_blocks = num_blocks
+        self.block_size = block_size
+        self._free: List[int] = list(range(num_blocks - 1, -1, -1))
+        self._tables: Dict[int, List[int]] = {}
+        self._lens: Dict[int, int] = {}
+
+    # ------------------------------------------------------------ lifecycle
+
+    def add_seq(self, seq_id: int) -> None:
+        if seq_id in self._tables:
+            raise KeyError(f"seq {seq_id} already exists")
+        self._tables[seq_id] = []
+        self._lens[seq_id] = 0
+
+    def free_seq(self, seq_id: int) -> None:
+        blocks = self._tables.pop(seq_id, None)
+        self._lens.pop(seq_id, None)
+        if blocks:
+            self._free.extend(reversed(blocks))
+
+    def has_seq(self, seq_id: int) -> bool:
+        return seq_id in self._tables
+
+    # ----------------------------------------------------------- allocation
+
+    def can_append(self, seq_id: int, n_tokens: int) -> bool:
+        need = self._blocks_needed(seq_id, n_tokens)
+        return need <= len(self._free)
+
+    def _blocks_needed(self, seq_id: int, n_tokens: int) -> int:
+        cur_len = self._lens[seq_id]
+        have = len(self._tables[seq_id])
+        need_total = (cur_len + n_tokens + self.block_size - 1) // self.block_size
+        return max(0, need_total - have)
+
+    def append_tokens(self, seq_id: int, n_tokens: int) -> List[int]:
+        """Reserve space for n new tokens; returns the flat slot index for
+        each token (block_id * block_size + offset)."""
+        need = self._blocks_needed(seq_id, n_tokens)
+        if need > len(self._free):
+            raise OutOfBlocksError(
+                f"seq {seq_id}: need {need} blocks, {len(self._free)} free"
+            )
+        table = self._tables[seq_id]
+        for _ in range(need):
+            table.append(self._free.pop())
+        slots = []
+        start = self._lens[seq_id]
+        for i in range(n_tokens):
+            pos = start + i
+            slots.append(table[pos // self.block_size] * self.block_size + pos % self.block_size)
+        self._lens[seq_id] = start + n_tokens
+        return slots
+
+    # -------------------------------------------------------------- queries
+
+    def block_table(self, seq_id: int) -> List[int]:
+        return list(self._tables[seq_id])
+
+    def seq_len(self, seq_id: int) -> int:
+        return self._lens[seq_id]
+
+    @property
+    def free_blocks(self) -> int:
+        return len(self._free)
+
+    @property
+    def used_blocks(self) -> int:
+        return self.num_blocks - len(self._free)
+
+    def occupancy(self) -> float:
+        return self.used_blocks / max(1, self.num_blocks)
+
+
+def make_block_manager(num_blocks: int, block_size: int, prefer_native: bool = True):
+    """Return the C++ block manager when the extension is built, else the
+    python one (CPU tests)."""
+    if prefer_native:
+        try:
+            from .. import _C  # built by build_ext.py
+
+            return _C.BlockManager(num_blocks, block_size)
+        except ImportError:
+            pass
+    return PyBlockManager(num_blocks, block_size)

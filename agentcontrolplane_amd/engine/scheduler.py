@@ -1,0 +1,263 @@
+"""Continuous-batching scheduler over the paged KV pool.
+
+Every step admits (a) one decode token for every running sequence and (b)
+prefill chunks from the waiting queue under a token budget
+(``max_prefill_tokens``), producing one FlatBatch — the Task reconcilers'
+concurrent LLM turns all land in the same forward pass (the "continuously
+batch every Pending/ToolCallsPending Task" requirement of the north star).
+
+KV pressure: when the pool cannot fit the next decode token for every
+running sequence, the youngest sequences are preempted — their blocks are
+freed and they re-enter the waiting queue with prompt = original prompt +
+tokens generated so far (recompute-style preemption; the context window in
+Task status stays the durable source of truth, the KV pool is a derived
+cache).
+"""
+from __future__ import annotations
+
+import dataclasses
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from .batch import FlatBatch, SeqMeta
+from .config import EngineConfig
+from .grammar import ToolCallGrammar
+from .kv import OutOfBlocksError
+from .request import InferenceRequest
+
+WAITING, PREFILL, DECODE, FINISHED = "waiting", "prefill", "decode", "finished"
+
+
+class Sequence:
+    __slots__ = (
+        "seq_id", "request", "prompt_ids", "output_ids", "num_processed",
+        "state", "grammar", "arrival",
+    )
+
+    def __init__(self, seq_id: int, request: InferenceRequest):
+        self.seq_id = seq_id
+        self.request = request
+        self.prompt_ids = list(request.prompt_ids)
+        self.output_ids: List[int] = []
+        self.num_processed = 0      # prompt tokens already in the KV cache
+        self.state = WAITING
+        self.grammar: Optional[ToolCallGrammar] = None
+        self.arrival = request.submit_time
+        if request.constrained:
+            names = [t["function"]["name"] for t in request.tools]
+            self.grammar = ToolCallGrammar(names)
+
+    @property
+    def total_len(self) -> int:
+        return len(self.prompt_ids) + len(self.output_ids)
+
+
+@dataclasses.dataclass
+class SchedulerOutput:
+    batch: Optional[FlatBatch]
+    preempted: List[Sequence]
+
+
+class Scheduler:
+    def __init__(self, config: EngineConfig, block_manager, device):
+        self.cfg = config
+        self.bm = block_manager
+        self.device = torch.device(device)
+        self.waiting: List[Sequence] = []
+        self.running: List[Sequence] = []   # in admission order (oldest first)
+        self._seq_counter = 0
+
+    def add_request(self, request: InferenceRequest) -> Sequence:
+        self._seq_counter += 1
+        seq = Sequence(self._seq_counter, request)
+        request.seq = seq
+        self.waiting.append(seq)
+        return seq
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def num_running(self) -> int:
+        return len(self.running)
+
+    # ---------------------------------------------------------------- step
+
+    def schedule(self) -> SchedulerOutput:
+        """Build the next step's FlatBatch."""
+        preempted: List[Sequence] = []
+        bs = self.cfg.kv_block_size
+
+        def blocks_for(cur_len: int, have_blocks: int, n_tokens: int) -> int:
+            total = (cur_len + n_tokens + bs - 1) // bs
+            return max(0, total - have_blocks)
+
+        def seq_blocks_for(s: Sequence, n_tokens: int) -> int:
+            return blocks_for(self.bm.seq_len(s.seq_id), len(self.bm.block_table(s.seq_id)), n_tokens)
+
+        # 1) reserve one decode token per running decode sequence; preempt
+        #    the youngest until the reservations fit the pool
+        decode_seqs = [s for s in self.running if s.state == DECODE]
+        while decode_seqs:
+            decode_need = sum(seq_blocks_for(s, 1) for s in decode_seqs)
+            if decode_need <= self.bm.free_blocks:
+                break
+            victim = decode_seqs.pop()  # youngest
+            self.running.remove(victim)
+            self._preempt(victim)
+            preempted.append(victim)
+        avail = self.bm.free_blocks - sum(seq_blocks_for(s, 1) for s in decode_seqs)
+
+        def fit_chunk(cur_len: int, have_blocks: int, want: int) -> int:
+            """Largest chunk ≤ want that fits in `avail` new blocks."""
+            slack = have_blocks * bs - cur_len  # room in the last partial block
+            return min(want, slack + avail * bs)
+
+        # 2) admit prefill chunks under the token budget and block budget
+        budget = self.cfg.max_prefill_tokens
+        prefills: List[Tuple[Sequence, int]] = []
+        # continue partially-prefilled running sequences first
+        for s in self.running:
+            if s.state == PREFILL and budget > 0:
+                want = min(len(s.prompt_ids) - s.num_processed, budget)
+                cur = self.bm.seq_len(s.seq_id)
+                have = len(self.bm.block_table(s.seq_id))
+                chunk = fit_chunk(cur, have, want)
+                if chunk > 0:
+                    prefills.append((s, chunk))
+                    budget -= chunk
+                    avail -= blocks_for(cur, have, chunk)
+        # then admit waiting sequences
+        while self.waiting and budget > 0 and len(self.running) < self.cfg.max_batch_size:
+            s = self.waiting[0]
+            want = min(len(s.prompt_ids), budget)
+            chunk = fit_chunk(0, 0, want)
+            if chunk <= 0:
+                break
+            self.waiting.pop(0)
+            self.bm.add_seq(s.seq_id)
+            s.state = PREFILL
+            self.running.append(s)
+            prefills.append((s, chunk))
+            budget -= chunk
+            avail -= blocks_for(0, 0, chunk)
+
+        decode_seqs = [s for s in self.running if s.state == DECODE]
+        if not prefills and not decode_seqs:
+            return SchedulerOutput(batch=None, preempted=preempted)
+
+        # 3) materialize the flat batch
+        token_ids: List[int] = []
+        positions: List[int] = []
+        slot_mapping: List[int] = []
+        prefill_metas: List[SeqMeta] = []
+        logit_rows: List[int] = []
+        sample_seq_ids: List[int] = []
+        row = 0
+        for s, chunk in prefills:
+            start = s.num_processed
+            toks = s.prompt_ids[start : start + chunk]
+            slots = self.bm.append_tokens(s.seq_id, chunk)
+            token_ids.extend(toks)
+            positions.extend(range(start, start + chunk))
+            slot_mapping.extend(slots)
+            s.num_processed += chunk
+            done = s.num_processed == len(s.prompt_ids)
+            prefill_metas.append(
+                SeqMeta(
+                    seq_id=s.seq_id,
+                    query_len=chunk,
+                    seq_len=s.num_processed,
+                    ctx_len=start,
+                    block_table=list(self.bm.block_table(s.seq_id)),
+                    needs_logits=done,
+                )
+            )
+            if done:
+                logit_rows.append(row + chunk - 1)
+                sample_seq_ids.append(s.seq_id)
+                s.state = DECODE
+            row += chunk
+        num_prefill_tokens = row
+
+        decode_block_tables = None
+        decode_seq_lens = None
+        decode_ids: List[int] = []
+        if decode_seqs:
+            max_blocks = 0
+            tables = []
+            lens = []
+            for s in decode_seqs:
+                tok = s.output_ids[-1] if s.output_ids else s.prompt_ids[-1]
+                pos = s.total_len - 1
+                slots = self.bm.append_tokens(s.seq_id, 1)
+                token_ids.append(tok)
+                positions.append(pos)
+                slot_mapping.extend(slots)
+                t = self.bm.block_table(s.seq_id)
+                tables.append(list(t))
+                lens.append(pos + 1)
+                max_blocks = max(max_blocks, len(t))
+                logit_rows.append(row)
+                sample_seq_ids.append(s.seq_id)
+                decode_ids.append(s.seq_id)
+                row += 1
+            padded = [t + [0] * (max_blocks - len(t)) for t in tables]
+            decode_block_tables = torch.tensor(padded, device=self.device, dtype=torch.long)
+            decode_seq_lens = torch.tensor(lens, device=self.device, dtype=torch.long)
+
+        batch = FlatBatch(
+            token_ids=torch.tensor(token_ids, device=self.device, dtype=torch.long),
+            positions=torch.tensor(positions, device=self.device, dtype=torch.long),
+            slot_mapping=torch.tensor(slot_mapping, device=self.device, dtype=torch.long),
+            prefills=prefill_metas,
+            num_prefill_tokens=num_prefill_tokens,
+            decode_seq_ids=decode_ids,
+            decode_block_tables=decode_block_tables,
+            decode_seq_lens=decode_seq_lens,
+            logit_rows=torch.tensor(logit_rows, device=self.device, dtype=torch.long),
+            sample_seq_ids=sample_seq_ids,
+        )
+        return SchedulerOutput(batch=batch, preempted=preempted)
+
+    # ------------------------------------------------------------ commit
+
+    def seq_by_id(self, seq_id: int) -> Optional[Sequence]:
+        for s in self.running:
+            if s.seq_id == seq_id:
+                return s
+        return None
+
+    def append_sampled(self, seq: Sequence, token: int) -> None:
+        seq.output_ids.append(token)
+        seq.request.output_ids.append(token)
+
+    def finish_seq(self, seq: Sequence, reason: str) -> None:
+        seq.state = FINISHED
+        if seq in self.running:
+            self.running.remove(seq)
+        if self.bm.has_seq(seq.seq_id):
+            self.bm.free_seq(seq.seq_id)
+        seq.request.finish(reason)
+
+    def _preempt(self, seq: Sequence) -> None:
+        """Recompute-style preemption: blocks freed, prompt grows to include
+        generated tokens, back to the head of the waiting queue."""
+        if self.bm.has_seq(seq.seq_id):
+            self.bm.free_seq(seq.seq_id)
+        seq.prompt_ids = seq.prompt_ids + seq.output_ids
+        # keep outputs in the request — generation continues from here
+        seq.output_ids = []
+        seq.num_processed = 0
+        seq.state = WAITING
+        self.waiting.insert(0, seq)
+
+    def abort(self, seq: Sequence, err: BaseException) -> None:
+        seq.state = FINISHED
+        if seq in self.running:
+            self.running.remove(seq)
+        if seq in self.waiting:
+            self.waiting.remove(seq)
+        if self.bm.has_seq(seq.seq_id):
+            self.bm.free_seq(seq.seq_id)
+        seq.request.fail(err)

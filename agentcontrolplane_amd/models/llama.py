@@ -1,0 +1,200 @@
+"""Llama-family model over the MI355X op set.
+
+Architecture (Llama-3): RMSNorm → fused-QKV proj → RoPE (rotate-half,
+θ=500000) → GQA attention on the paged KV cache → O proj → RMSNorm →
+SwiGLU MLP, pre-norm residuals, untied LM head.
+
+Execution model: a flat FlatBatch per step (mixed prefill chunks + decode
+tokens).  GEMMs go through torch.nn.functional.linear (hipBLASLt on ROCm);
+everything between the GEMMs — rmsnorm, rope+KV-scatter, attention over the
+paged cache, SwiGLU — is the op set in ``agentcontrolplane_amd.ops`` (HIP
+kernels on a GPU, fp32 torch references on CPU).
+
+Tensor parallelism: head-parallel attention + column/row-parallel MLP;
+rank r holds heads [r·Hq/w, (r+1)·Hq/w) and MLP columns likewise; the
+all-reduce after o-proj and down-proj is issued by the caller (engine
+worker) so communication can overlap with the next layer's norm.
+"""
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+import torch.nn.functional as F
+
+from .. import ops
+from ..engine.batch import FlatBatch
+from ..engine.config import EngineConfig, ModelConfig
+from ..ops.reference import build_cos_sin
+
+
+def _dtype_of(cfg: ModelConfig):
+    return {"bfloat16": torch.bfloat16, "float16": torch.float16, "float32": torch.float32}[
+        cfg.dtype
+    ]
+
+
+class LlamaLayerWeights:
+    __slots__ = ("input_norm", "qkv", "o", "post_norm", "gate_up", "down")
+
+
+class LlamaForCausalLM:
+    def __init__(
+        self,
+        cfg: ModelConfig,
+        ecfg: EngineConfig,
+        device: str,
+        tp_rank: int = 0,
+        tp_world: int = 1,
+    ):
+        self.cfg = cfg
+        self.ecfg = ecfg
+        self.device = torch.device(device)
+        self.dtype = _dtype_of(cfg)
+        self.tp_rank = tp_rank
+        self.tp_world = tp_world
+        assert cfg.num_heads % tp_world == 0 and cfg.num_kv_heads % tp_world == 0, (
+            "TP degree must divide head counts"
+        )
+        self.n_heads = cfg.num_heads // tp_world
+        self.n_kv_heads = cfg.num_kv_heads // tp_world
+        self.inter = cfg.intermediate_size // tp_world
+        self.head_dim = cfg.head_dim
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        self.layers: List[LlamaLayerWeights] = []
+        self.embed: Optional[torch.Tensor] = None
+        self.lm_head: Optional[torch.Tensor] = None
+        self.final_norm: Optional[torch.Tensor] = None
+        self.cos_sin = build_cos_sin(
+            cfg.max_position, cfg.head_dim, cfg.rope_theta, self.device
+        )
+        # per-layer paged KV caches, allocated by allocate_kv_cache
+        self.k_caches: List[torch.Tensor] = []
+        self.v_caches: List[torch.Tensor] = []
+        # optional TP all-reduce hook installed by the parallel engine
+        self.all_reduce = None
+
+    # ------------------------------------------------------------- weights
+
+    def random_init(self, seed: int = 0) -> None:
+        """Random-init weights of the real shapes (synthetic bench: no
+        network for checkpoints — BASELINE.md).  Init in manageable chunks
+        directly on the device."""
+        g = torch.Generator(device="cpu").manual_seed(seed + self.tp_rank)
+        cfg = self.cfg
+        h = cfg.hidden_size
+        std = 0.02
+
+        def randw(*shape):
+            w = torch.empty(shape, dtype=self.dtype, device=self.device)
+            tmp = torch.empty(shape, dtype=torch.float16 if self.dtype != torch.float32 else torch.float32)
+            tmp.normal_(0.0, std, generator=g)
+            w.copy_(tmp)
+            return w
+
+        self.embed = randw(cfg.vocab_size, h)
+        self.lm_head = self.embed if cfg.tie_embeddings else randw(cfg.vocab_size, h)
+        self.final_norm = torch.ones(h, dtype=self.dtype, device=self.device)
+        qd = self.n_heads * self.head_dim
+        kvd = self.n_kv_heads * self.head_dim
+        for _ in range(cfg.num_layers):
+            lw = LlamaLayerWeights()
+            lw.input_norm = torch.ones(h, dtype=self.dtype, device=self.device)
+            lw.qkv = randw(qd + 2 * kvd, h)
+            lw.o = randw(h, qd)
+            lw.post_norm = torch.ones(h, dtype=self.dtype, device=self.device)
+            lw.gate_up = randw(2 * self.inter, h)
+            lw.down = randw(h, self.inter)
+            self.layers.append(lw)
+
+    def load_safetensors(self, path: str) -> None:  # pragma: no cover - no checkpoints here
+        """Checkpoint loading seam (safetensors shard dir, HF llama naming).
+        Unused in this deployment (no network for weights) but kept so a real
+        checkpoint drops in."""
+        import json
+        import os
+
+        from safetensors.torch import load_file
+
+        raise NotImplementedError("checkpoint loading lands with a reachable weights source")
+
+    # ------------------------------------------------------------ KV cache
+
+    def allocate_kv_cache(self, num_blocks: int, block_size: int) -> None:
+        shape = (num_blocks, block_size, self.n_kv_heads, self.head_dim)
+        self.k_caches = [
+            torch.zeros(shape, dtype=self.dtype, device=self.device)
+            for _ in range(self.cfg.num_layers)
+        ]
+        self.v_caches = [
+            torch.zeros(shape, dtype=self.dtype, device=self.device)
+            for _ in range(self.cfg.num_layers)
+        ]
+
+    def kv_block_bytes(self, block_size: int) -> int:
+        return (
+            2 * self.cfg.num_layers * block_size * self.n_kv_heads * self.head_dim
+            * self.dtype.itemsize
+        )
+
+    # -------------------------------------------------------------- forward
+
+    def _attention(self, layer_idx: int, q, batch: FlatBatch):
+        """q: [N, Hq, D] post-rope; returns [N, Hq, D]."""
+        kc, vc = self.k_caches[layer_idx], self.v_caches[layer_idx]
+        outs = []
+        row = 0
+        for m in batch.prefills:
+            outs.append(
+                ops.attention_prefill(
+                    q[row : row + m.query_len], kc, vc, m.block_table, m.seq_len,
+                    m.ctx_len, self.scale,
+                )
+            )
+            row += m.query_len
+        if batch.num_decode:
+            outs.append(
+                ops.attention_decode_batch(
+                    q[row:], kc, vc, batch.decode_block_tables, batch.decode_seq_lens,
+                    self.scale,
+                )
+            )
+        return torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
+
+    def forward(self, batch: FlatBatch) -> torch.Tensor:
+        """Returns logits [len(batch.sample_seq_ids), vocab]."""
+        cfg = self.cfg
+        N = batch.num_tokens
+        x = F.embedding(batch.token_ids, self.embed)
+        residual = None
+        qd = self.n_heads * self.head_dim
+        kvd = self.n_kv_heads * self.head_dim
+        for li, lw in enumerate(self.layers):
+            if residual is None:
+                residual = x
+                normed = ops.rmsnorm(x, lw.input_norm, cfg.rms_eps)
+            else:
+                normed, residual = ops.fused_add_rmsnorm(x, residual, lw.input_norm, cfg.rms_eps)
+            qkv = F.linear(normed, lw.qkv)
+            q = qkv[:, :qd].view(N, self.n_heads, self.head_dim)
+            k = qkv[:, qd : qd + kvd].view(N, self.n_kv_heads, self.head_dim)
+            v = qkv[:, qd + kvd :].view(N, self.n_kv_heads, self.head_dim)
+            q, k = ops.rope_and_cache(
+                q.contiguous(), k.contiguous(), v.contiguous(),
+                batch.positions, batch.slot_mapping,
+                self.k_caches[li], self.v_caches[li], self.cos_sin,
+            )
+            attn = self._attention(li, q, batch)
+            o = F.linear(attn.reshape(N, qd), lw.o)
+            if self.all_reduce is not None:
+                o = self.all_reduce(o)
+            normed2, residual = ops.fused_add_rmsnorm(o, residual, lw.post_norm, cfg.rms_eps)
+            gate_up = F.linear(normed2, lw.gate_up)
+            mlp = F.linear(ops.swiglu(gate_up), lw.down)
+            if self.all_reduce is not None:
+                mlp = self.all_reduce(mlp)
+            x = mlp
+        normed, _ = ops.fused_add_rmsnorm(x, residual, self.final_norm, cfg.rms_eps)
+        sel = normed[batch.logit_rows]
+        return F.linear(sel, self.lm_head).float()

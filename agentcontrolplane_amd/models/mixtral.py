@@ -1,0 +1,117 @@
+"""Mixtral-family (sparse MoE) model.
+
+Same attention stack as Llama; the dense SwiGLU MLP becomes a top-2-of-8
+expert mixture: router linear → softmax over experts → top-k weights
+renormalized → weighted sum of the chosen experts' SwiGLU outputs.
+
+Single-GPU execution groups tokens by expert and runs one gate_up/down GEMM
+pair per expert with >0 tokens.  Under expert parallelism
+(parallel/moe.py), experts are sharded across ranks and token routing goes
+through an RCCL all-to-all over xGMI.
+"""
+from __future__ import annotations
+
+from typing import List
+
+import torch
+import torch.nn.functional as F
+
+from .. import ops
+from ..engine.config import EngineConfig, ModelConfig
+from .llama import LlamaForCausalLM
+
+
+class MixtralForCausalLM(LlamaForCausalLM):
+    def __init__(self, cfg: ModelConfig, ecfg: EngineConfig, device: str,
+                 tp_rank: int = 0, tp_world: int = 1):
+        super().__init__(cfg, ecfg, device, tp_rank, tp_world)
+        self.num_experts = cfg.num_experts
+        self.top_k = cfg.num_experts_per_tok
+        self.routers: List[torch.Tensor] = []
+        self.expert_gate_up: List[torch.Tensor] = []   # per layer: [E, 2I, H]
+        self.expert_down: List[torch.Tensor] = []      # per layer: [E, H, I]
+        # expert-parallel dispatch hook installed by parallel/moe.py
+        self.moe_dispatch = None
+
+    def random_init(self, seed: int = 0) -> None:
+        super().random_init(seed)
+        g = torch.Generator(device="cpu").manual_seed(seed + 1000 + self.tp_rank)
+        h = self.cfg.hidden_size
+        std = 0.02
+
+        def randw(*shape):
+            w = torch.empty(shape, dtype=self.dtype, device=self.device)
+            tmp = torch.empty(
+                shape, dtype=torch.float16 if self.dtype != torch.float32 else torch.float32
+            )
+            tmp.normal_(0.0, std, generator=g)
+            w.copy_(tmp)
+            return w
+
+        self.routers = []
+        self.expert_gate_up = []
+        self.expert_down = []
+        for lw in self.layers:
+            # the dense MLP weights are replaced per-expert; free them
+            lw.gate_up = None
+            lw.down = None
+            self.routers.append(randw(self.num_experts, h))
+            self.expert_gate_up.append(randw(self.num_experts, 2 * self.inter, h))
+            self.expert_down.append(randw(self.num_experts, h, self.inter))
+
+    def _moe_mlp(self, li: int, x: torch.Tensor) -> torch.Tensor:
+        """x: [N, H] → [N, H] via top-k expert mixture."""
+        router_logits = F.linear(x, self.routers[li]).float()       # [N, E]
+        probs = torch.softmax(router_logits, dim=-1)
+        topw, topi = torch.topk(probs, self.top_k, dim=-1)          # [N, k]
+        topw = topw / topw.sum(dim=-1, keepdim=True)
+        if self.moe_dispatch is not None:
+            return self.moe_dispatch(li, x, topi, topw.to(x.dtype))
+        out = torch.zeros_like(x)
+        flat_expert = topi.reshape(-1)                              # [N*k]
+        flat_rows = (
+            torch.arange(x.shape[0], device=x.device).repeat_interleave(self.top_k)
+        )
+        flat_w = topw.reshape(-1).to(x.dtype)
+        for e in range(self.num_experts):
+            sel = (flat_expert == e).nonzero(as_tuple=True)[0]
+            if sel.numel() == 0:
+                continue
+            rows = flat_rows[sel]
+            xe = x[rows]
+            gate_up = F.linear(xe, self.expert_gate_up[li][e])
+            ye = F.linear(ops.swiglu(gate_up), self.expert_down[li][e])
+            out.index_add_(0, rows, ye * flat_w[sel].unsqueeze(-1))
+        return out
+
+    def forward(self, batch) -> torch.Tensor:
+        cfg = self.cfg
+        N = batch.num_tokens
+        x = F.embedding(batch.token_ids, self.embed)
+        residual = None
+        qd = self.n_heads * self.head_dim
+        kvd = self.n_kv_heads * self.head_dim
+        for li, lw in enumerate(self.layers):
+            if residual is None:
+                residual = x
+                normed = ops.rmsnorm(x, lw.input_norm, cfg.rms_eps)
+            else:
+                normed, residual = ops.fused_add_rmsnorm(x, residual, lw.input_norm, cfg.rms_eps)
+            qkv = F.linear(normed, lw.qkv)
+            q = qkv[:, :qd].view(N, self.n_heads, self.head_dim)
+            k = qkv[:, qd : qd + kvd].view(N, self.n_kv_heads, self.head_dim)
+            v = qkv[:, qd + kvd :].view(N, self.n_kv_heads, self.head_dim)
+            q, k = ops.rope_and_cache(
+                q.contiguous(), k.contiguous(), v.contiguous(),
+                batch.positions, batch.slot_mapping,
+                self.k_caches[li], self.v_caches[li], self.cos_sin,
+            )
+            attn = self._attention(li, q, batch)
+            o = F.linear(attn.reshape(N, qd), lw.o)
+            if self.all_reduce is not None:
+                o = self.all_reduce(o)
+            normed2, residual = ops.fused_add_rmsnorm(o, residual, lw.post_norm, cfg.rms_eps)
+            x = self._moe_mlp(li, normed2)
+        normed, _ = ops.fused_add_rmsnorm(x, residual, self.final_norm, cfg.rms_eps)
+        sel = normed[batch.logit_rows]
+        return F.linear(sel, self.lm_head).float()

@@ -1,0 +1,80 @@
+"""Op dispatch.
+
+Two implementations of every engine op:
+
+- ``reference`` — plain PyTorch fp32, runs on CPU.  The numerics oracle for
+  every HIP kernel test and the compute path for CPU-only engine tests.
+- ``hip`` — hand-written CDNA4 HIP kernels (csrc/*.hip) through the in-tree
+  ``_C`` extension.  The ONLY path on a GPU: if a tensor is on cuda and the
+  extension is missing, ops raise instead of silently falling back, so a GPU
+  run that passes is guaranteed to have run the native kernels.
+"""
+from __future__ import annotations
+
+import torch
+
+_HIP = None
+_HIP_ERR: Exception | None = None
+
+
+def _hip():
+    global _HIP, _HIP_ERR
+    if _HIP is None and _HIP_ERR is None:
+        try:
+            from . import hip as m
+
+            _HIP = m
+        except Exception as e:  # pragma: no cover
+            _HIP_ERR = e
+    if _HIP is None:
+        raise RuntimeError(
+            f"HIP ops extension not available on a GPU tensor path: {_HIP_ERR}\n"
+            "Build it with: python build_ext.py (PYTORCH_ROCM_ARCH=gfx950)"
+        )
+    return _HIP
+
+
+def _impl(t: torch.Tensor):
+    if t.is_cuda:
+        return _hip()
+    from . import reference as m
+
+    return m
+
+
+def rmsnorm(x, weight, eps: float = 1e-5):
+    return _impl(x).rmsnorm(x, weight, eps)
+
+
+def fused_add_rmsnorm(x, residual, weight, eps: float = 1e-5):
+    """returns (normed, new_residual) where new_residual = x + residual."""
+    return _impl(x).fused_add_rmsnorm(x, residual, weight, eps)
+
+
+def rope_and_cache(q, k, v, positions, slot_mapping, k_cache, v_cache, cos_sin):
+    """Apply rotary embedding to q,k in place and scatter k,v into the paged
+    cache at slot_mapping.  q:[N, Hq, D] k/v:[N, Hkv, D]."""
+    return _impl(q).rope_and_cache(q, k, v, positions, slot_mapping, k_cache, v_cache, cos_sin)
+
+
+def attention_prefill(q, k_cache, v_cache, block_table, seq_len, ctx_len, scale):
+    """Causal attention for one sequence's prefill chunk.
+    q: [Lq, Hq, D]; KV read from the paged cache; ctx_len = tokens already
+    cached before this chunk (the chunk's causal offset)."""
+    return _impl(q).attention_prefill(q, k_cache, v_cache, block_table, seq_len, ctx_len, scale)
+
+
+def attention_decode_batch(q, k_cache, v_cache, block_tables, seq_lens, scale):
+    """One-token-per-sequence paged attention.  q: [B, Hq, D]."""
+    return _impl(q).attention_decode_batch(q, k_cache, v_cache, block_tables, seq_lens, scale)
+
+
+def swiglu(gate_up):
+    """gate_up: [N, 2*I] (gate | up interleaved as two halves) → [N, I]."""
+    return _impl(gate_up).swiglu(gate_up)
+
+
+def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None):
+    """Batch sampling: temperature/top-k/top-p with optional per-row boolean
+    vocab masks (constrained decoding).  Returns [B] long."""
+    return _impl(logits).softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask)
