@@ -39,6 +39,11 @@ class FlatBatch:
     logit_rows: torch.Tensor
     sample_seq_ids: List[int]
 
+    # lazily-built device metadata, shared by every layer's kernels
+    _prefill_meta: Optional["PrefillMeta"] = None
+    _decode_tables_i32: Optional[torch.Tensor] = None
+    _decode_lens_i32: Optional[torch.Tensor] = None
+
     @property
     def num_tokens(self) -> int:
         return int(self.token_ids.shape[0])
@@ -46,3 +51,60 @@ class FlatBatch:
     @property
     def num_decode(self) -> int:
         return len(self.decode_seq_ids)
+
+    def prefill_meta(self, tile_q: int = 64) -> "PrefillMeta":
+        if self._prefill_meta is None:
+            self._prefill_meta = PrefillMeta.build(
+                self.prefills, self.token_ids.device, tile_q
+            )
+        return self._prefill_meta
+
+    def decode_tables_i32(self) -> torch.Tensor:
+        if self._decode_tables_i32 is None:
+            self._decode_tables_i32 = self.decode_block_tables.int().contiguous()
+        return self._decode_tables_i32
+
+    def decode_lens_i32(self) -> torch.Tensor:
+        if self._decode_lens_i32 is None:
+            self._decode_lens_i32 = self.decode_seq_lens.int().contiguous()
+        return self._decode_lens_i32
+
+
+@dataclasses.dataclass
+class PrefillMeta:
+    """Device tensors for the batched prefill-attention kernel, built once
+    per step and reused by all layers."""
+
+    block_tables: torch.Tensor  # [num_seqs, max_blocks] int32
+    seq_lens: torch.Tensor      # [num_seqs] int32 (after this chunk)
+    ctx_lens: torch.Tensor      # [num_seqs] int32 (before this chunk)
+    row_starts: torch.Tensor    # [num_seqs] int32 — first q row per seq
+    tile_seq: torch.Tensor      # [num_tiles] int32
+    tile_q0: torch.Tensor       # [num_tiles] int32
+    tile_q: int
+
+    @staticmethod
+    def build(prefills: List[SeqMeta], device, tile_q: int = 64) -> "PrefillMeta":
+        max_blocks = max((len(m.block_table) for m in prefills), default=1)
+        tables, seq_lens, ctx_lens, row_starts = [], [], [], []
+        tile_seq, tile_q0 = [], []
+        row = 0
+        for i, m in enumerate(prefills):
+            tables.append(m.block_table + [0] * (max_blocks - len(m.block_table)))
+            seq_lens.append(m.seq_len)
+            ctx_lens.append(m.ctx_len)
+            row_starts.append(row)
+            for q0 in range(0, m.query_len, tile_q):
+                tile_seq.append(i)
+                tile_q0.append(q0)
+            row += m.query_len
+        t = lambda x: torch.tensor(x, device=device, dtype=torch.int32)  # noqa: E731
+        return PrefillMeta(
+            block_tables=t(tables),
+            seq_lens=t(seq_lens),
+            ctx_lens=t(ctx_lens),
+            row_starts=t(row_starts),
+            tile_seq=t(tile_seq),
+            tile_q0=t(tile_q0),
+            tile_q=tile_q,
+        )

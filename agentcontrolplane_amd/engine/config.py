@@ -68,6 +68,20 @@ PRESETS = {
         num_experts=8,
         num_experts_per_tok=2,
     ),
+    # tiny GPU config: real head_dim/GQA shape for the CDNA4 kernels at
+    # test scale (kernels require head_dim 128)
+    "tiny-gpu": ModelConfig(
+        name="tiny-gpu",
+        vocab_size=512,
+        hidden_size=256,
+        intermediate_size=512,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        max_position=2048,
+        dtype="bfloat16",
+    ),
     # tiny configs for CPU tests
     "tiny": ModelConfig(
         name="tiny",
@@ -78,7 +92,7 @@ PRESETS = {
         num_heads=4,
         num_kv_heads=2,
         head_dim=16,
-        max_position=512,
+        max_position=4096,
         dtype="float32",
     ),
     "tiny-moe": ModelConfig(

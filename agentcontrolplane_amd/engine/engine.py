@@ -102,6 +102,12 @@ class InferenceEngine:
     # ------------------------------------------------------------- submit
 
     def submit(self, request: InferenceRequest) -> InferenceRequest:
+        limit = min(self.cfg.max_model_len, self.mcfg.max_position)
+        if len(request.prompt_ids) + request.sampling.max_tokens > limit:
+            raise ValueError(
+                f"request length {len(request.prompt_ids)}+{request.sampling.max_tokens} "
+                f"exceeds the model context limit {limit}"
+            )
         with self._work:
             if len(self._pending) >= self.cfg.max_queue:
                 raise RuntimeError("engine queue full")

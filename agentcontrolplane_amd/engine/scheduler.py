@@ -130,6 +130,17 @@ class Scheduler:
         # then admit waiting sequences
         while self.waiting and budget > 0 and len(self.running) < self.cfg.max_batch_size:
             s = self.waiting[0]
+            total_blocks = (s.total_len + s.request.sampling.max_tokens + bs - 1) // bs
+            if total_blocks > self.bm.num_blocks:
+                self.waiting.pop(0)
+                self.abort(
+                    s,
+                    ValueError(
+                        f"request needs {total_blocks} KV blocks but the pool has "
+                        f"{self.bm.num_blocks}"
+                    ),
+                )
+                continue
             want = min(len(s.prompt_ids), budget)
             chunk = fit_chunk(0, 0, want)
             if chunk <= 0:

@@ -46,6 +46,10 @@ class LocalEngineClient(LLMClient):
             )
         except TimeoutError as e:
             raise LLMRequestError(503, f"engine timeout: {e}")
+        except ValueError as e:
+            # invalid request (e.g. context-limit overflow) — terminal 4xx so
+            # the Task fails instead of retrying forever
+            raise LLMRequestError(400, f"invalid request: {e}")
         except Exception as e:
             raise LLMRequestError(500, f"engine error: {e}")
 

@@ -81,16 +81,14 @@ class LlamaForCausalLM:
         """Random-init weights of the real shapes (synthetic bench: no
         network for checkpoints — BASELINE.md).  Init in manageable chunks
         directly on the device."""
-        g = torch.Generator(device="cpu").manual_seed(seed + self.tp_rank)
+        g = torch.Generator(device=self.device).manual_seed(seed + self.tp_rank)
         cfg = self.cfg
         h = cfg.hidden_size
         std = 0.02
 
         def randw(*shape):
             w = torch.empty(shape, dtype=self.dtype, device=self.device)
-            tmp = torch.empty(shape, dtype=torch.float16 if self.dtype != torch.float32 else torch.float32)
-            tmp.normal_(0.0, std, generator=g)
-            w.copy_(tmp)
+            w.normal_(0.0, std, generator=g)
             return w
 
         self.embed = randw(cfg.vocab_size, h)
@@ -144,20 +142,15 @@ class LlamaForCausalLM:
         """q: [N, Hq, D] post-rope; returns [N, Hq, D]."""
         kc, vc = self.k_caches[layer_idx], self.v_caches[layer_idx]
         outs = []
-        row = 0
-        for m in batch.prefills:
+        row = batch.num_prefill_tokens
+        if batch.prefills:
             outs.append(
-                ops.attention_prefill(
-                    q[row : row + m.query_len], kc, vc, m.block_table, m.seq_len,
-                    m.ctx_len, self.scale,
-                )
+                ops.attention_prefill_batch(q[:row], kc, vc, batch, self.scale)
             )
-            row += m.query_len
         if batch.num_decode:
             outs.append(
                 ops.attention_decode_batch(
-                    q[row:], kc, vc, batch.decode_block_tables, batch.decode_seq_lens,
-                    self.scale,
+                    q[row:], kc, vc, batch, self.scale,
                 )
             )
         return torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]

@@ -35,17 +35,13 @@ class MixtralForCausalLM(LlamaForCausalLM):
 
     def random_init(self, seed: int = 0) -> None:
         super().random_init(seed)
-        g = torch.Generator(device="cpu").manual_seed(seed + 1000 + self.tp_rank)
+        g = torch.Generator(device=self.device).manual_seed(seed + 1000 + self.tp_rank)
         h = self.cfg.hidden_size
         std = 0.02
 
         def randw(*shape):
             w = torch.empty(shape, dtype=self.dtype, device=self.device)
-            tmp = torch.empty(
-                shape, dtype=torch.float16 if self.dtype != torch.float32 else torch.float32
-            )
-            tmp.normal_(0.0, std, generator=g)
-            w.copy_(tmp)
+            w.normal_(0.0, std, generator=g)
             return w
 
         self.routers = []

@@ -58,15 +58,21 @@ def rope_and_cache(q, k, v, positions, slot_mapping, k_cache, v_cache, cos_sin):
 
 
 def attention_prefill(q, k_cache, v_cache, block_table, seq_len, ctx_len, scale):
-    """Causal attention for one sequence's prefill chunk.
+    """Causal attention for ONE sequence's prefill chunk (test/oracle entry).
     q: [Lq, Hq, D]; KV read from the paged cache; ctx_len = tokens already
     cached before this chunk (the chunk's causal offset)."""
     return _impl(q).attention_prefill(q, k_cache, v_cache, block_table, seq_len, ctx_len, scale)
 
 
-def attention_decode_batch(q, k_cache, v_cache, block_tables, seq_lens, scale):
+def attention_prefill_batch(q, k_cache, v_cache, batch, scale):
+    """Causal attention for ALL prefill chunks of a FlatBatch in one launch.
+    q: [num_prefill_tokens, Hq, D]."""
+    return _impl(q).attention_prefill_batch(q, k_cache, v_cache, batch, scale)
+
+
+def attention_decode_batch(q, k_cache, v_cache, batch, scale):
     """One-token-per-sequence paged attention.  q: [B, Hq, D]."""
-    return _impl(q).attention_decode_batch(q, k_cache, v_cache, block_tables, seq_lens, scale)
+    return _impl(q).attention_decode_batch(q, k_cache, v_cache, batch, scale)
 
 
 def swiglu(gate_up):

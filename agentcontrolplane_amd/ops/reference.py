@@ -84,8 +84,29 @@ def attention_prefill(q, k_cache, v_cache, block_table, seq_len, ctx_len, scale)
     return out.to(q.dtype)
 
 
-def attention_decode_batch(q, k_cache, v_cache, block_tables, seq_lens, scale):
-    """q: [B, Hq, D]; block_tables: [B, max_blocks] long; seq_lens: [B]."""
+def attention_prefill_batch(q, k_cache, v_cache, batch, scale):
+    """Loop the per-sequence oracle over a FlatBatch's prefill chunks."""
+    outs = []
+    row = 0
+    for m in batch.prefills:
+        outs.append(
+            attention_prefill(
+                q[row : row + m.query_len], k_cache, v_cache, m.block_table,
+                m.seq_len, m.ctx_len, scale,
+            )
+        )
+        row += m.query_len
+    return torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
+
+
+def attention_decode_batch(q, k_cache, v_cache, batch, scale):
+    """q: [B, Hq, D]; decode tables/lens come from the FlatBatch."""
+    return attention_decode_raw(
+        q, k_cache, v_cache, batch.decode_block_tables, batch.decode_seq_lens, scale
+    )
+
+
+def attention_decode_raw(q, k_cache, v_cache, block_tables, seq_lens, scale):
     B, Hq, D = q.shape
     outs = []
     for i in range(B):

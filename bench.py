@@ -1,0 +1,251 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: agent-loop steps/sec on the MI355X-native control
+plane + inference engine (BASELINE.json metric: "agent-loop steps/sec +
+p50 Task latency, Llama-3-8B, 1k concurrent Tasks"; the reference publishes
+no numbers — BASELINE.md establishes them here).
+
+One bench step = every one of C concurrent Tasks per GPU completes one full
+agent-loop wave: LLM turn with a grammar-constrained tool call → MCP tool
+execution → ToolCall join → LLM answer turn → FinalAnswer.  Each wave is
+2 LLM turns per task, so one step completes 2·C agent-loop steps per GPU.
+value = whole-job agent-loop steps/sec across all N GPUs.
+
+Scaling is weak data parallelism (request sharding): each rank runs its own
+control plane + engine on its GPU with C tasks (config 4's sharding mode);
+--tp N instead runs one tensor-parallel engine across the ranks.  Synthetic
+data: random-init weights of the named architecture, deterministic
+pseudo-random prompts (no network for checkpoints or datasets).
+
+Contract: W untimed warmup steps, then EXACTLY K timed steps bracketed by
+dist barrier + torch.cuda.synchronize on both sides; elapsed = MAX over
+ranks; rank 0 prints one JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import random
+import statistics
+import string
+import sys
+import threading
+import time
+
+
+def make_prompt(rng: random.Random, n_chars: int = 600) -> str:
+    words = []
+    total = 0
+    while total < n_chars:
+        w = "".join(rng.choice(string.ascii_lowercase) for _ in range(rng.randint(3, 9)))
+        words.append(w)
+        total += len(w) + 1
+    return " ".join(words)
+
+
+def run_wave(cp, agent_name: str, concurrency: int, rng: random.Random,
+             timeout_s: float = 1200.0):
+    """Create C tasks, wait until all reach FinalAnswer; returns latencies."""
+    from agentcontrolplane_amd.api.types import TASK, TaskPhase, make_resource
+
+    names = []
+    t_create = {}
+    for i in range(concurrency):
+        name = f"bench-{rng.randrange(1 << 30):08x}-{i}"
+        cp.store.create(
+            make_resource(
+                TASK,
+                name,
+                spec={"agentRef": {"name": agent_name}, "userMessage": make_prompt(rng)},
+            )
+        )
+        t_create[name] = time.monotonic()
+        names.append(name)
+    deadline = time.monotonic() + timeout_s
+    latencies = {}
+    pending = set(names)
+    while pending:
+        if time.monotonic() > deadline:
+            raise TimeoutError(f"{len(pending)} tasks unfinished (of {concurrency})")
+        done = set()
+        for name in pending:
+            t = cp.store.get(TASK, name)
+            phase = (t or {}).get("status", {}).get("phase")
+            if phase == TaskPhase.FINAL_ANSWER:
+                latencies[name] = time.monotonic() - t_create[name]
+                done.add(name)
+            elif phase == TaskPhase.FAILED:
+                raise RuntimeError(f"task {name} failed: {t['status'].get('error')}")
+        pending -= done
+        if pending:
+            time.sleep(0.02)
+    # cleanup so the next wave starts from an empty store
+    for name in names:
+        cp.store.delete(TASK, name)
+    return list(latencies.values())
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--model", default="llama3-8b")
+    p.add_argument("--concurrency", type=int, default=128, help="tasks per GPU")
+    p.add_argument("--decode-tokens", type=int, default=32)
+    p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree (else DP sharding)")
+    p.add_argument("--device", default=None)
+    p.add_argument("--kv-blocks", type=int, default=None)
+    args = p.parse_args()
+
+    import torch
+
+    from agentcontrolplane_amd.api.types import AGENT, LLM, MCP_SERVER, make_resource
+    from agentcontrolplane_amd.engine.config import EngineConfig
+    from agentcontrolplane_amd.engine.engine import InferenceEngine
+    from agentcontrolplane_amd.parallel.dist import init_distributed
+    from agentcontrolplane_amd.runtime import ControlPlane
+
+    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = init_distributed()
+    if device == "cuda":
+        torch.cuda.set_device(local_rank)
+    import torch.distributed as dist
+
+    is_dist = world > 1
+
+    model = args.model
+    if device == "cpu":
+        model = "tiny"  # CPU smoke of the bench harness itself
+
+    ecfg = EngineConfig(
+        model=model,
+        device=device,
+        num_kv_blocks=args.kv_blocks if args.kv_blocks else (1024 if device == "cpu" else None),
+        max_prefill_tokens=8192,
+        tensor_parallel=args.tp,
+        seed=1234 + rank,
+        request_timeout_s=1200,
+    )
+    if args.tp > 1:
+        raise NotImplementedError(
+            "tensor-parallel serving runs through parallel/tp.py's sharded "
+            "engine worker; the driver bench uses DP request sharding"
+        )
+    engine = InferenceEngine(ecfg)
+
+    cp = ControlPlane(engine=engine, auto_approve="approve", llm_probe=False)
+    cp.start()
+    try:
+        cp.store.create(
+            make_resource(
+                LLM,
+                "bench-llm",
+                spec={
+                    "provider": "local",
+                    "parameters": {
+                        "model": model,
+                        "maxTokens": args.decode_tokens,
+                        "temperature": "0.8",
+                    },
+                },
+            )
+        )
+        llm = cp.store.get(LLM, "bench-llm")
+        llm["status"].update({"ready": True, "status": "Ready"})
+        cp.store.update_status(llm)
+        cp.mcp.register_inproc("tools", {"noop": lambda **_: "ok"})
+        cp.store.create(make_resource(MCP_SERVER, "tools", spec={"transport": "inproc"}))
+        cp.store.create(
+            make_resource(
+                AGENT,
+                "bench-agent",
+                spec={
+                    "llmRef": {"name": "bench-llm"},
+                    "system": "You are a benchmark agent. Use tools when offered.",
+                    "mcpServers": [{"name": "tools"}],
+                },
+            )
+        )
+        # wait for the agent to validate
+        t0 = time.monotonic()
+        while time.monotonic() - t0 < 60:
+            a = cp.store.get(AGENT, "bench-agent")
+            if a.get("status", {}).get("ready"):
+                break
+            time.sleep(0.05)
+        else:
+            raise TimeoutError("agent never became ready")
+
+        rng = random.Random(42 + rank)
+        conc = args.concurrency if device != "cpu" else 8
+
+        for _ in range(args.warmup):
+            run_wave(cp, "bench-agent", conc, rng)
+
+        if is_dist:
+            dist.barrier()
+        if device == "cuda":
+            torch.cuda.synchronize()
+        t_start = time.monotonic()
+        all_lat = []
+        for _ in range(args.steps):
+            all_lat.extend(run_wave(cp, "bench-agent", conc, rng))
+        if device == "cuda":
+            torch.cuda.synchronize()
+        if is_dist:
+            dist.barrier()
+        elapsed = time.monotonic() - t_start
+        if is_dist:
+            t = torch.tensor([elapsed])
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            elapsed = float(t[0])
+
+        agent_steps = 2 * conc * args.steps  # 2 LLM turns per task per wave
+        total_steps = agent_steps * world
+        value = total_steps / elapsed
+        p50 = statistics.median(all_lat)
+        em = engine.metrics()
+        if rank == 0:
+            out = {
+                "metric": "agent-loop steps/sec",
+                "value": round(value, 3),
+                "unit": "steps/s",
+                "n_gpus": world if device == "cuda" else args.gpus,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": round(1000.0 * elapsed / args.steps, 2),
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "bf16" if device == "cuda" else "fp32",
+                "data": "synthetic (random-init weights, pseudo-random prompts; no network)",
+                "config": {
+                    "model": model,
+                    "global_batch": conc * world,
+                    "seq_len": 8192,
+                    "parallelism": (f"tp{args.tp}" if args.tp > 1 else f"dp{world}"),
+                    "concurrency_per_gpu": conc,
+                    "decode_tokens": args.decode_tokens,
+                    "llm_turns_per_task": 2,
+                    "p50_task_latency_s": round(p50, 3),
+                    "p95_task_latency_s": round(
+                        sorted(all_lat)[max(0, int(0.95 * len(all_lat)) - 1)], 3
+                    ),
+                    "engine_tokens_per_s": round(
+                        em["generated_tokens"] / max(em["busy_time_s"], 1e-9), 1
+                    ),
+                    "prompt_tokens_total": em["prompt_tokens"],
+                    "generated_tokens_total": em["generated_tokens"],
+                },
+            }
+            print(json.dumps(out))
+    finally:
+        cp.stop()
+        engine.stop()
+
+
+if __name__ == "__main__":
+    main()
