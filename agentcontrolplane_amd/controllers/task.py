@@ -503,8 +503,9 @@ class TaskReconciler(Reconciler):
         tool_calls = self.store.list(
             TOOL_CALL, ns, label_selector={LABEL_TASK: name, LABEL_TCREQ: req_id}
         )
-        if not tool_calls:
-            return Result(requeue_after=DEFAULT_REQUEUE)
+        # an empty set counts as all-completed (reference checkToolCalls
+        # semantics, state_machine.go:304-317): a turn with neither content
+        # nor tool calls loops straight back to the LLM
         pending = [
             tc
             for tc in tool_calls

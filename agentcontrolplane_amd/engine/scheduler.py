@@ -46,7 +46,11 @@ class Sequence:
         self.arrival = request.submit_time
         if request.constrained:
             names = [t["function"]["name"] for t in request.tools]
-            self.grammar = ToolCallGrammar(names)
+            # steer the arguments object to close before the token budget:
+            # the grammar's closing mode needs ~20 tokens of slack for the
+            # name/scaffolding plus the shortest legal unwind
+            max_args = max(16, request.sampling.max_tokens - 24 - max(len(n) for n in names))
+            self.grammar = ToolCallGrammar(names, max_args_len=max_args)
 
     @property
     def total_len(self) -> int:
