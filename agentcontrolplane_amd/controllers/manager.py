@@ -151,6 +151,7 @@ class ControllerManager:
         self._error_backoff: Dict[Tuple[str, str, str], float] = {}
 
     def register(self, rec: Reconciler) -> None:
+        rec._manager = self  # back-ref for async completion requeues
         self._reconcilers.append(rec)
         self._queues[rec.kind] = _WorkQueue()
 

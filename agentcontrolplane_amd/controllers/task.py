@@ -87,6 +87,11 @@ class TaskReconciler(Reconciler):
         self.tracer = get_tracer()
         self._mutexes: Dict[Tuple[str, str], threading.Lock] = {}
         self._mutex_guard = threading.Lock()
+        # in-flight async LLM turns, keyed by (ns, name): workers submit to
+        # the engine and return; the completion callback requeues the task
+        # (SURVEY.md §7 step 5 — no one-blocking-call-per-reconcile)
+        self._inflight: Dict[Tuple[str, str], dict] = {}
+        self._manager = None  # back-ref set by ControllerManager.register
 
     # ToolCall events requeue the parent task via the task label
     def map_owned(self, ev):
@@ -303,13 +308,31 @@ class TaskReconciler(Reconciler):
 
     def _send_llm_request_locked(self, task) -> Result:
         # re-read under the lock: another pod may have advanced the phase
-        fresh = self.store.get(TASK, task["metadata"]["name"], task["metadata"].get("namespace", "default"))
+        name = task["metadata"]["name"]
+        ns = task["metadata"].get("namespace", "default")
+        fresh = self.store.get(TASK, name, ns)
         if fresh is None or fresh["status"].get("phase") != TaskPhase.READY_FOR_LLM:
+            self._inflight.pop((ns, name), None)
             return Result()
         task = fresh
         status = task["status"]
         spec = task.get("spec", {})
-        ns = task["metadata"].get("namespace", "default")
+        key = (ns, name)
+        marker = len(status.get("contextWindow", []))
+
+        # a completed async turn? process it under the lock/lease
+        entry = self._inflight.get(key)
+        if entry is not None:
+            if entry["marker"] != marker:
+                # context moved on while the turn was in flight — drop it
+                self._inflight.pop(key, None)
+            elif not entry["done"]:
+                return Result()  # completion callback will requeue
+            else:
+                self._inflight.pop(key, None)
+                if entry["error"] is not None:
+                    return self._handle_llm_error(task, entry["error"])
+                return self._process_llm_response(task, entry["output"], entry["tools"])
 
         agent = self.store.get(AGENT, (spec.get("agentRef") or {}).get("name", ""), ns)
         if agent is None:
@@ -355,18 +378,46 @@ class TaskReconciler(Reconciler):
         )
         try:
             client = self.factory.create_client(llm, api_key)
-            output = client.send_request(messages, tools)
-            llm_span.set_status("OK")
         except Exception as e:
             llm_span.record_error(e)
             llm_span.set_status("ERROR", str(e))
             llm_span.end()
             return self._handle_llm_error(task, e)
-        finally:
-            if llm_span.end_ns == 0:
-                llm_span.end()
 
-        return self._process_llm_response(task, output, tools)
+        entry = {
+            "marker": marker,
+            "done": False,
+            "output": None,
+            "error": None,
+            "tools": tools,
+            "span": llm_span,
+        }
+        self._inflight[key] = entry
+
+        def _on_turn(output, error, _entry=entry, _key=key):
+            span = _entry["span"]
+            if error is not None:
+                span.record_error(error)
+                span.set_status("ERROR", str(error))
+            else:
+                span.set_status("OK")
+            span.end()
+            _entry["output"] = output
+            _entry["error"] = error
+            _entry["done"] = True
+            if self._manager is not None:
+                self._manager.enqueue(TASK, _key[1], _key[0])
+
+        client.send_request_async(messages, tools, _on_turn)
+
+        # synchronous clients (mock, remote stub) complete inline — finish
+        # the turn in this reconcile instead of an extra queue hop
+        if entry["done"] and self._inflight.get(key) is entry:
+            self._inflight.pop(key, None)
+            if entry["error"] is not None:
+                return self._handle_llm_error(task, entry["error"])
+            return self._process_llm_response(task, entry["output"], entry["tools"])
+        return Result()
 
     def _handle_llm_error(self, task, err) -> Result:
         """4xx terminal vs retry-in-place (state_machine.go:733-789)."""

@@ -62,6 +62,9 @@ class InferenceEngine:
             "preemptions": 0,
             "requests_completed": 0,
             "busy_time_s": 0.0,
+            "sched_time_s": 0.0,
+            "compute_time_s": 0.0,
+            "sample_time_s": 0.0,
         }
         self._start_time = time.monotonic()
         if start:
@@ -128,14 +131,7 @@ class InferenceEngine:
         req.wait(timeout or self.cfg.request_timeout_s)
         return req
 
-    def chat(
-        self,
-        messages: List[Dict[str, Any]],
-        tools: Optional[List[Dict[str, Any]]] = None,
-        sampling: Optional[SamplingParams] = None,
-    ) -> ChatResult:
-        """The LLMClient.send_request counterpart."""
-        t0 = time.monotonic()
+    def _build_request(self, messages, tools, sampling) -> InferenceRequest:
         sampling = sampling or SamplingParams()
         tools = tools or []
         prompt_ids = self.tokenizer.render_chat(messages, tools)
@@ -150,11 +146,12 @@ class InferenceEngine:
                 constrained = not has_tool_result
             elif choice == "required":
                 constrained = True
-        req = InferenceRequest(prompt_ids, sampling, constrained, tools)
-        self.submit(req)
-        out_ids = req.wait(self.cfg.request_timeout_s)
+        return InferenceRequest(prompt_ids, sampling, constrained, tools)
+
+    def _result_of(self, req: InferenceRequest, t0: float) -> ChatResult:
+        out_ids = req.output_ids
         result = ChatResult(
-            prompt_tokens=len(prompt_ids),
+            prompt_tokens=len(req.prompt_ids),
             completion_tokens=len(out_ids),
             finish_reason=req.finish_reason,
             latency_s=time.monotonic() - t0,
@@ -171,6 +168,44 @@ class InferenceEngine:
         else:
             result.text = self.tokenizer.decode(out_ids)
         return result
+
+    def chat(
+        self,
+        messages: List[Dict[str, Any]],
+        tools: Optional[List[Dict[str, Any]]] = None,
+        sampling: Optional[SamplingParams] = None,
+    ) -> ChatResult:
+        """The LLMClient.send_request counterpart (blocking)."""
+        t0 = time.monotonic()
+        req = self._build_request(messages, tools, sampling)
+        self.submit(req)
+        req.wait(self.cfg.request_timeout_s)
+        return self._result_of(req, t0)
+
+    def chat_async(self, messages, tools, sampling, callback) -> InferenceRequest:
+        """Submit a chat turn; ``callback(result, error)`` fires from the
+        engine thread on completion.  This is the serving path: reconciler
+        workers never block on a turn."""
+        t0 = time.monotonic()
+        req = self._build_request(messages, tools, sampling)
+
+        def on_complete(r: InferenceRequest) -> None:
+            try:
+                if r.error is not None:
+                    callback(None, r.error)
+                else:
+                    callback(self._result_of(r, t0), None)
+            except Exception:  # noqa: BLE001 — callback bugs must not kill the loop
+                import traceback
+
+                traceback.print_exc()
+
+        req.on_complete = on_complete
+        try:
+            self.submit(req)
+        except Exception as e:
+            callback(None, e)
+        return req
 
     # ------------------------------------------------------------ the loop
 
@@ -198,13 +233,20 @@ class InferenceEngine:
     def step(self) -> bool:
         """One scheduler round + forward + sampling.  Returns True if any
         work ran."""
+        t0 = time.monotonic()
         out = self.scheduler.schedule()
+        t1 = time.monotonic()
         self._m["preemptions"] += len(out.preempted)
         if out.batch is None:
             return False
         batch = out.batch
         logits = self.model.forward(batch)
+        t2 = time.monotonic()
         self._sample_and_commit(batch, logits)
+        t3 = time.monotonic()
+        self._m["sched_time_s"] += t1 - t0
+        self._m["compute_time_s"] += t2 - t1   # launch-side; sample syncs
+        self._m["sample_time_s"] += t3 - t2
         self._m["steps"] += 1
         self._m["prompt_tokens"] += batch.num_prefill_tokens
         return True

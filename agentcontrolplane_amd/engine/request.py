@@ -54,16 +54,27 @@ class InferenceRequest:
         self.finish_reason = "stop"
         self._event = threading.Event()
         self._error: Optional[BaseException] = None
+        # invoked (from the engine thread) when the request completes; used
+        # by the async serving path to requeue the owning Task reconcile
+        self.on_complete = None
         # scheduler state
         self.seq = None  # assigned by the scheduler
 
     def finish(self, reason: str) -> None:
         self.finish_reason = reason
         self._event.set()
+        if self.on_complete is not None:
+            self.on_complete(self)
 
     def fail(self, err: BaseException) -> None:
         self._error = err
         self._event.set()
+        if self.on_complete is not None:
+            self.on_complete(self)
+
+    @property
+    def error(self) -> Optional[BaseException]:
+        return self._error
 
     def wait(self, timeout: Optional[float] = None) -> List[int]:
         if not self._event.wait(timeout):

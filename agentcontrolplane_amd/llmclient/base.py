@@ -57,10 +57,23 @@ class Tool:
 
 
 class LLMClient:
-    """Uniform chat-completion interface (llm_client.go:11-14)."""
+    """Uniform chat-completion interface (llm_client.go:11-14).
+
+    ``send_request`` blocks; ``send_request_async`` delivers the assistant
+    Message (or error) through a callback so reconciler workers never block
+    on a turn — engine-backed clients override it with a true async submit,
+    the default runs synchronously in the calling thread (mock/remote)."""
 
     def send_request(self, messages: List[Message], tools: List[Tool]) -> Message:
         raise NotImplementedError
+
+    def send_request_async(self, messages, tools, callback) -> None:
+        try:
+            msg = self.send_request(messages, tools)
+        except Exception as e:  # noqa: BLE001 — delivered to the callback
+            callback(None, e)
+            return
+        callback(msg, None)
 
 
 def tool_from_contact_channel(channel: Dict[str, Any]) -> Tool:

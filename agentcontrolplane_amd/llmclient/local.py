@@ -30,29 +30,55 @@ class LocalEngineClient(LLMClient):
         self.top_p = float(params.get("topP") or 1.0)
         self.top_k = int(params.get("topK") or 0)
 
-    def send_request(self, messages: List[Message], tools: List[Tool]) -> Message:
+    def _sampling(self):
         from ..engine.request import SamplingParams
 
+        return SamplingParams(
+            max_tokens=int(self.max_tokens),
+            temperature=self.temperature,
+            top_p=self.top_p,
+            top_k=self.top_k,
+        )
+
+    @staticmethod
+    def _map_error(e: BaseException) -> LLMRequestError:
+        if isinstance(e, TimeoutError):
+            return LLMRequestError(503, f"engine timeout: {e}")
+        if isinstance(e, ValueError):
+            # invalid request (e.g. context-limit overflow) — terminal 4xx so
+            # the Task fails instead of retrying forever
+            return LLMRequestError(400, f"invalid request: {e}")
+        return LLMRequestError(500, f"engine error: {e}")
+
+    def send_request(self, messages: List[Message], tools: List[Tool]) -> Message:
         try:
             result = self.engine.chat(
                 messages=[m.to_dict() for m in messages],
                 tools=[t.to_dict() for t in tools],
-                sampling=SamplingParams(
-                    max_tokens=int(self.max_tokens),
-                    temperature=self.temperature,
-                    top_p=self.top_p,
-                    top_k=self.top_k,
-                ),
+                sampling=self._sampling(),
             )
-        except TimeoutError as e:
-            raise LLMRequestError(503, f"engine timeout: {e}")
-        except ValueError as e:
-            # invalid request (e.g. context-limit overflow) — terminal 4xx so
-            # the Task fails instead of retrying forever
-            raise LLMRequestError(400, f"invalid request: {e}")
         except Exception as e:
-            raise LLMRequestError(500, f"engine error: {e}")
+            raise self._map_error(e)
+        return self._to_message(result)
 
+    def send_request_async(self, messages, tools, callback) -> None:
+        """Non-blocking submit into the continuous-batching engine; the
+        callback fires from the engine thread at turn completion."""
+
+        def _cb(result, error):
+            if error is not None:
+                callback(None, self._map_error(error))
+            else:
+                callback(self._to_message(result), None)
+
+        self.engine.chat_async(
+            [m.to_dict() for m in messages],
+            [t.to_dict() for t in tools],
+            self._sampling(),
+            _cb,
+        )
+
+    def _to_message(self, result) -> Message:
         msg = Message(role="assistant")
         if result.tool_calls:
             msg.tool_calls = [
