@@ -239,6 +239,13 @@ def main() -> None:
                     ),
                     "prompt_tokens_total": em["prompt_tokens"],
                     "generated_tokens_total": em["generated_tokens"],
+                    "engine_steps": em["steps"],
+                    "engine_busy_s": round(em["busy_time_s"], 2),
+                    "engine_sched_s": round(em["sched_time_s"], 2),
+                    "engine_compute_s": round(em["compute_time_s"], 2),
+                    "engine_sample_s": round(em["sample_time_s"], 2),
+                    "kv_occupancy": round(em["kv_occupancy"], 4),
+                    "preemptions": em["preemptions"],
                 },
             }
             print(json.dumps(out))
