@@ -79,6 +79,21 @@ def run_wave(cp, agent_name: str, concurrency: int, rng: random.Random,
         pending -= done
         if pending:
             time.sleep(0.02)
+    # forensics: the wave is gated by its slowest task — dump its event
+    # timeline (stderr; the stdout JSON line stays clean)
+    slowest = max(latencies, key=latencies.get)
+    ls = sorted(latencies.values())
+    print(
+        f"[wave] n={len(ls)} p50={ls[len(ls)//2]:.2f}s "
+        f"p95={ls[max(0,int(0.95*len(ls))-1)]:.2f}s max={ls[-1]:.2f}s ({slowest})",
+        file=sys.stderr,
+    )
+    if ls[-1] > 3 * ls[len(ls) // 2]:
+        evs = cp.store.events_for(slowest)
+        t_prev = None
+        for e in evs:
+            ts = e.get("lastTimestamp", "")
+            print(f"[straggler] {ts} {e['reason']}: {e['message'][:80]}", file=sys.stderr)
     # cleanup so the next wave starts from an empty store
     for name in names:
         cp.store.delete(TASK, name)
