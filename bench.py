@@ -48,6 +48,7 @@ def run_wave(cp, agent_name: str, concurrency: int, rng: random.Random,
     """Create C tasks, wait until all reach FinalAnswer; returns latencies."""
     from agentcontrolplane_amd.api.types import TASK, TaskPhase, make_resource
 
+    t_wave0 = time.monotonic()
     names = []
     t_create = {}
     for i in range(concurrency):
@@ -61,6 +62,7 @@ def run_wave(cp, agent_name: str, concurrency: int, rng: random.Random,
         )
         t_create[name] = time.monotonic()
         names.append(name)
+    t_created = time.monotonic()
     deadline = time.monotonic() + timeout_s
     latencies = {}
     pending = set(names)
@@ -81,12 +83,15 @@ def run_wave(cp, agent_name: str, concurrency: int, rng: random.Random,
             time.sleep(0.02)
     # forensics: the wave is gated by its slowest task — dump its event
     # timeline (stderr; the stdout JSON line stays clean)
+    t_polled = time.monotonic()
     slowest = max(latencies, key=latencies.get)
     ls = sorted(latencies.values())
     print(
-        f"[wave] n={len(ls)} p50={ls[len(ls)//2]:.2f}s "
+        f"[wave] n={len(ls)} create={t_created - t_wave0:.2f}s "
+        f"poll={t_polled - t_created:.2f}s p50={ls[len(ls)//2]:.2f}s "
         f"p95={ls[max(0,int(0.95*len(ls))-1)]:.2f}s max={ls[-1]:.2f}s ({slowest})",
         file=sys.stderr,
+        flush=True,
     )
     if ls[-1] > 3 * ls[len(ls) // 2]:
         evs = cp.store.events_for(slowest)
@@ -97,6 +102,7 @@ def run_wave(cp, agent_name: str, concurrency: int, rng: random.Random,
     # cleanup so the next wave starts from an empty store
     for name in names:
         cp.store.delete(TASK, name)
+    print(f"[wave] cleanup={time.monotonic() - t_polled:.2f}s", file=sys.stderr, flush=True)
     return list(latencies.values())
 
 
