@@ -100,8 +100,24 @@ def run_wave(cp, agent_name: str, concurrency: int, rng: random.Random,
             ts = e.get("lastTimestamp", "")
             print(f"[straggler] {ts} {e['reason']}: {e['message'][:80]}", file=sys.stderr)
     # cleanup so the next wave starts from an empty store
+    profile = os.environ.get("ACP_BENCH_PROFILE") == "1"
+    if profile:
+        import cProfile
+        import io
+        import pstats
+
+        pr = cProfile.Profile()
+        pr.enable()
     for name in names:
         cp.store.delete(TASK, name)
+    if profile:
+        pr.disable()
+        st = pstats.Stats(pr)
+        st.sort_stats("cumulative")
+        buf = io.StringIO()
+        st.stream = buf
+        st.print_stats(15)
+        print(buf.getvalue()[:4000], file=sys.stderr, flush=True)
     print(f"[wave] cleanup={time.monotonic() - t_polled:.2f}s", file=sys.stderr, flush=True)
     return list(latencies.values())
 
