@@ -244,6 +244,12 @@ class ResourceStore:
             if cur is None:
                 raise NotFoundError(f"{kind} {ns}/{name} not found")
             self._check_rv(cur, obj)
+            if which == "status" and cur.get("status") == obj.get("status", {}):
+                # no-op status write: like the apiserver, identical content
+                # does not bump resourceVersion or emit a watch event —
+                # without this, reconcilers that rewrite unchanged status
+                # self-trigger forever through their own MODIFIED events
+                return copy.deepcopy(cur)
             self._rv += 1
             if which == "status":
                 cur["status"] = copy.deepcopy(obj.get("status", {}))
@@ -311,6 +317,20 @@ class ResourceStore:
         (the reference has ~40 recorder.Event call sites, e.g.
         task/state_machine.go:224,628,731)."""
         m = involved.get("metadata", {})
+        # k8s-style event aggregation: an identical (object, reason, message)
+        # event bumps count/lastTimestamp instead of creating a new object
+        with self._lock:
+            ns_bucket = self._data.get(EVENT, {}).get(m.get("namespace", "default"), {})
+            for ev in ns_bucket.values():
+                if (
+                    ev.get("involvedObject", {}).get("name") == m.get("name")
+                    and ev.get("reason") == reason
+                    and ev.get("message") == message
+                    and ev.get("type") == event_type
+                ):
+                    ev["count"] = int(ev.get("count", 1)) + 1
+                    ev["lastTimestamp"] = now_iso()
+                    return
         name = f'{m.get("name", "obj")}.{self._rv + 1}'
         ev = {
             "apiVersion": "v1",
@@ -327,6 +347,7 @@ class ResourceStore:
             "type": event_type,
             "reason": reason,
             "message": message,
+            "count": 1,
             "lastTimestamp": now_iso(),
         }
         with self._lock:
