@@ -54,10 +54,12 @@ class FlatBatch:
 
     def prefill_meta(self, tile_q: int = 64) -> "PrefillMeta":
         if self._prefill_meta is None:
-            self._prefill_meta = PrefillMeta.build(
+            self._prefill_meta = {}
+        if tile_q not in self._prefill_meta:
+            self._prefill_meta[tile_q] = PrefillMeta.build(
                 self.prefills, self.token_ids.device, tile_q
             )
-        return self._prefill_meta
+        return self._prefill_meta[tile_q]
 
     def decode_tables_i32(self) -> torch.Tensor:
         if self._decode_tables_i32 is None:

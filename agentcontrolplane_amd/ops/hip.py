@@ -47,13 +47,25 @@ def attention_prefill(q, k_cache, v_cache, block_table, seq_len, ctx_len, scale)
     return attention_prefill_batch(q, k_cache, v_cache, batch, scale)
 
 
+import os
+
+_PREFILL_IMPL = os.environ.get("ACP_PREFILL_IMPL", "mfma")  # mfma | v0
+
+
 def attention_prefill_batch(q, k_cache, v_cache, batch, scale):
-    meta = batch.prefill_meta(tile_q=64)
     out = torch.empty_like(q)
-    _C.prefill_attn(
-        out, q, k_cache, v_cache, meta.block_tables, meta.seq_lens, meta.ctx_lens,
-        meta.row_starts, meta.tile_seq, meta.tile_q0, scale,
-    )
+    if _PREFILL_IMPL == "mfma":
+        meta = batch.prefill_meta(tile_q=128)
+        _C.prefill_attn_mfma(
+            out, q, k_cache, v_cache, meta.block_tables, meta.seq_lens, meta.ctx_lens,
+            meta.row_starts, meta.tile_seq, meta.tile_q0, scale,
+        )
+    else:
+        meta = batch.prefill_meta(tile_q=64)
+        _C.prefill_attn(
+            out, q, k_cache, v_cache, meta.block_tables, meta.seq_lens, meta.ctx_lens,
+            meta.row_starts, meta.tile_seq, meta.tile_q0, scale,
+        )
     return out
 
 

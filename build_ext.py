@@ -28,6 +28,8 @@ def build(verbose: bool = False) -> str:
         os.path.join(CSRC, "rope_cache.hip"),
         os.path.join(CSRC, "decode_attn.hip"),
         os.path.join(CSRC, "prefill_attn.hip"),
+        os.path.join(CSRC, "prefill_attn_mfma.hip"),
+        os.path.join(CSRC, "mfma_probe.hip"),
         os.path.join(CSRC, "sampling.hip"),
     ]
     module = cpp_extension.load(

@@ -25,6 +25,10 @@ void launch_prefill_attn(void*, const void*, const void*, const void*, const int
                          float, int, int, int, int, int, int, hipStream_t);
 void launch_sample(int64_t*, const float*, const float*, const int64_t*, const float*,
                    const float*, const uint8_t*, int, int, hipStream_t);
+void launch_prefill_attn_mfma(void*, const void*, const void*, const void*, const int*,
+                              const int*, const int*, const int*, const int*, const int*,
+                              float, int, int, int, int, int, int, hipStream_t);
+void launch_mfma_probe(float*, const void*, const void*, hipStream_t);
 }
 
 #define CHECK_CUDA(x) TORCH_CHECK(x.is_cuda(), #x " must be a GPU tensor")
@@ -102,11 +106,11 @@ static void decode_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cach
                      max_blocks, kv_block, cur_stream());
 }
 
-static void prefill_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
-                         torch::Tensor v_cache, torch::Tensor block_tables,
-                         torch::Tensor seq_lens, torch::Tensor ctx_lens,
-                         torch::Tensor row_starts, torch::Tensor tile_seq,
-                         torch::Tensor tile_q0, double scale) {
+static void prefill_attn_impl(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                              torch::Tensor v_cache, torch::Tensor block_tables,
+                              torch::Tensor seq_lens, torch::Tensor ctx_lens,
+                              torch::Tensor row_starts, torch::Tensor tile_seq,
+                              torch::Tensor tile_q0, double scale, bool mfma) {
   CHECK_CUDA(q); CHECK_CONT(q); CHECK_BF16(q); CHECK_CONT(out);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt);
   const int Hq = q.size(1), D = q.size(2);
@@ -115,12 +119,38 @@ static void prefill_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cac
   const int max_blocks = block_tables.size(1);
   const int num_tiles = tile_seq.size(0);
   TORCH_CHECK(D == 128, "prefill kernel requires head_dim 128");
-  launch_prefill_attn(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
-                      v_cache.data_ptr(), block_tables.data_ptr<int>(),
-                      seq_lens.data_ptr<int>(), ctx_lens.data_ptr<int>(),
-                      row_starts.data_ptr<int>(), tile_seq.data_ptr<int>(),
-                      tile_q0.data_ptr<int>(), (float)scale, num_tiles, Hq, Hkv, D,
-                      max_blocks, kv_block, cur_stream());
+  auto fn = mfma ? launch_prefill_attn_mfma : launch_prefill_attn;
+  fn(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+     v_cache.data_ptr(), block_tables.data_ptr<int>(),
+     seq_lens.data_ptr<int>(), ctx_lens.data_ptr<int>(),
+     row_starts.data_ptr<int>(), tile_seq.data_ptr<int>(),
+     tile_q0.data_ptr<int>(), (float)scale, num_tiles, Hq, Hkv, D,
+     max_blocks, kv_block, cur_stream());
+}
+
+static void prefill_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                         torch::Tensor v_cache, torch::Tensor block_tables,
+                         torch::Tensor seq_lens, torch::Tensor ctx_lens,
+                         torch::Tensor row_starts, torch::Tensor tile_seq,
+                         torch::Tensor tile_q0, double scale) {
+  prefill_attn_impl(out, q, k_cache, v_cache, block_tables, seq_lens, ctx_lens,
+                    row_starts, tile_seq, tile_q0, scale, /*mfma=*/false);
+}
+
+static void prefill_attn_mfma(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                              torch::Tensor v_cache, torch::Tensor block_tables,
+                              torch::Tensor seq_lens, torch::Tensor ctx_lens,
+                              torch::Tensor row_starts, torch::Tensor tile_seq,
+                              torch::Tensor tile_q0, double scale) {
+  prefill_attn_impl(out, q, k_cache, v_cache, block_tables, seq_lens, ctx_lens,
+                    row_starts, tile_seq, tile_q0, scale, /*mfma=*/true);
+}
+
+static torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
+  CHECK_CUDA(a); CHECK_BF16(a); CHECK_CONT(a); CHECK_CONT(b);
+  auto d = torch::empty({32, 32}, a.options().dtype(at::kFloat));
+  launch_mfma_probe(d.data_ptr<float>(), a.data_ptr(), b.data_ptr(), cur_stream());
+  return d;
 }
 
 static void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
@@ -150,5 +180,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_cache", &rope_cache);
   m.def("decode_attn", &decode_attn);
   m.def("prefill_attn", &prefill_attn);
+  m.def("prefill_attn_mfma", &prefill_attn_mfma);
+  m.def("mfma_probe", &mfma_probe);
   m.def("sample", &sample);
 }

@@ -119,10 +119,30 @@ def test_decode_attn(G):
 
 
 @requires_gpu
-def test_prefill_attn_multi_seq():
+def test_mfma_fragment_layout_probe():
+    """The assumed A/B/D lane maps for v_mfma_f32_32x32x16_bf16 must
+    reproduce a plain matmul (asymmetric operands per guide G9/rule 16)."""
+    import torch
+
+    from agentcontrolplane_amd import _C
+
+    torch.manual_seed(7)
+    a = (torch.randn(32, 16) * 0.5).bfloat16().cuda()
+    b = (torch.randn(16, 32) * 0.5).bfloat16().cuda()
+    d = _C.mfma_probe(a, b)
+    want = a.cpu().float() @ b.cpu().float()
+    err = (d.cpu() - want).abs().max().item()
+    assert err < 0.05, f"fragment layout mismatch: max err {err}"
+
+
+@requires_gpu
+@pytest.mark.parametrize("impl", ["v0", "mfma"])
+def test_prefill_attn_multi_seq(impl, monkeypatch):
     """Three chunks with different ctx offsets in one batched launch."""
     from agentcontrolplane_amd.engine.batch import FlatBatch, SeqMeta
+    from agentcontrolplane_amd.ops import hip as hip_mod
 
+    monkeypatch.setattr(hip_mod, "_PREFILL_IMPL", impl)
     torch.manual_seed(2)
     Hq, Hkv, D, bs = 4, 2, 128, 16
     num_blocks = 128
