@@ -42,11 +42,27 @@ class InferenceEngine:
             self.model = create_model(self.mcfg, config, config.device)
             self.model.random_init(config.seed)
         num_blocks = self._size_kv_pool()
-        self.model.allocate_kv_cache(num_blocks, config.kv_block_size)
+        # +1 scratch block: hipGraph decode padding rows write their KV there
+        self.model.allocate_kv_cache(num_blocks + 1, config.kv_block_size)
         self.bm = make_block_manager(
             num_blocks, config.kv_block_size, prefer_native=True
         )
         self.scheduler = Scheduler(config, self.bm, self.device)
+        self.graph_runner = None
+        if self.device.type == "cuda" and not config.enforce_eager:
+            from .graphs import DecodeGraphRunner
+
+            max_blocks_per_seq = (
+                min(config.max_model_len, self.mcfg.max_position)
+                + config.kv_block_size - 1
+            ) // config.kv_block_size
+            self.graph_runner = DecodeGraphRunner(
+                self.model,
+                max_batch=min(config.max_batch_size, 512),
+                max_blocks_per_seq=max_blocks_per_seq,
+                scratch_block=num_blocks,
+                kv_block_size=config.kv_block_size,
+            )
         self._gen = torch.Generator(device=self.device.type)
         self._gen.manual_seed(config.seed)
         self._lock = threading.Lock()
@@ -240,7 +256,13 @@ class InferenceEngine:
         if out.batch is None:
             return False
         batch = out.batch
-        logits = self.model.forward(batch)
+        logits = None
+        if self.graph_runner is not None and not batch.prefills:
+            logits = self.graph_runner.run(batch)
+            if logits is not None:
+                self._m["graph_steps"] = self._m.get("graph_steps", 0) + 1
+        if logits is None:
+            logits = self.model.forward(batch)
         t2 = time.monotonic()
         self._sample_and_commit(batch, logits)
         t3 = time.monotonic()

@@ -94,12 +94,16 @@ class MixtralForCausalLM(LlamaForCausalLM):
             else:
                 normed, residual = ops.fused_add_rmsnorm(x, residual, lw.input_norm, cfg.rms_eps)
             qkv = F.linear(normed, lw.qkv)
-            q = qkv[:, :qd].view(N, self.n_heads, self.head_dim)
-            k = qkv[:, qd : qd + kvd].view(N, self.n_kv_heads, self.head_dim)
-            v = qkv[:, qd + kvd :].view(N, self.n_kv_heads, self.head_dim)
+            row = qkv.stride(0)
+            q = qkv.as_strided((N, self.n_heads, self.head_dim), (row, self.head_dim, 1))
+            k = qkv.as_strided(
+                (N, self.n_kv_heads, self.head_dim), (row, self.head_dim, 1), qd
+            )
+            v = qkv.as_strided(
+                (N, self.n_kv_heads, self.head_dim), (row, self.head_dim, 1), qd + kvd
+            )
             q, k = ops.rope_and_cache(
-                q.contiguous(), k.contiguous(), v.contiguous(),
-                batch.positions, batch.slot_mapping,
+                q, k, v, batch.positions, batch.slot_mapping,
                 self.k_caches[li], self.v_caches[li], self.cos_sin,
             )
             attn = self._attention(li, q, batch)

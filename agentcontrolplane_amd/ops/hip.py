@@ -52,8 +52,12 @@ import os
 _PREFILL_IMPL = os.environ.get("ACP_PREFILL_IMPL", "mfma")  # mfma | v0
 
 
+def _attn_out(q):
+    return torch.empty(q.shape, dtype=q.dtype, device=q.device)
+
+
 def attention_prefill_batch(q, k_cache, v_cache, batch, scale):
-    out = torch.empty_like(q)
+    out = _attn_out(q)
     if _PREFILL_IMPL == "mfma":
         meta = batch.prefill_meta(tile_q=128)
         _C.prefill_attn_mfma(
@@ -70,7 +74,7 @@ def attention_prefill_batch(q, k_cache, v_cache, batch, scale):
 
 
 def attention_decode_batch(q, k_cache, v_cache, batch, scale):
-    out = torch.empty_like(q)
+    out = _attn_out(q)
     _C.decode_attn(
         out, q, k_cache, v_cache, batch.decode_tables_i32(), batch.decode_lens_i32(),
         scale,
@@ -79,7 +83,7 @@ def attention_decode_batch(q, k_cache, v_cache, batch, scale):
 
 
 def attention_decode_raw(q, k_cache, v_cache, block_tables, seq_lens, scale):
-    out = torch.empty_like(q)
+    out = _attn_out(q)
     _C.decode_attn(
         out, q, k_cache, v_cache, block_tables.int().contiguous(),
         seq_lens.int().contiguous(), scale,

@@ -17,23 +17,33 @@ void launch_fused_add_rmsnorm(void*, void*, const void*, const void*, float, int
 void launch_swiglu(void*, const void*, int64_t, int, hipStream_t);
 void launch_rope_cache(void*, void*, const void*, void*, void*, const int64_t*,
                        const int64_t*, const float*, int, int, int, int, int64_t,
-                       hipStream_t);
+                       int64_t, int64_t, hipStream_t);
 void launch_decode_attn(void*, const void*, const void*, const void*, const int*,
-                        const int*, float, int, int, int, int, int, int, hipStream_t);
+                        const int*, float, int, int, int, int, int, int, int64_t,
+                        hipStream_t);
 void launch_prefill_attn(void*, const void*, const void*, const void*, const int*,
                          const int*, const int*, const int*, const int*, const int*,
-                         float, int, int, int, int, int, int, hipStream_t);
+                         float, int, int, int, int, int, int, int64_t, hipStream_t);
 void launch_sample(int64_t*, const float*, const float*, const int64_t*, const float*,
                    const float*, const uint8_t*, int, int, hipStream_t);
 void launch_prefill_attn_mfma(void*, const void*, const void*, const void*, const int*,
                               const int*, const int*, const int*, const int*, const int*,
-                              float, int, int, int, int, int, int, hipStream_t);
+                              float, int, int, int, int, int, int, int64_t, hipStream_t);
 void launch_mfma_probe(float*, const void*, const void*, hipStream_t);
 }
 
 #define CHECK_CUDA(x) TORCH_CHECK(x.is_cuda(), #x " must be a GPU tensor")
 #define CHECK_CONT(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
 #define CHECK_BF16(x) TORCH_CHECK(x.scalar_type() == at::kBFloat16, #x " must be bf16")
+
+
+// q tensors may be strided views into the fused QKV GEMM output: dim-0
+// stride is free, heads/dims must be dense
+static int64_t q_token_stride(const torch::Tensor& t) {
+  TORCH_CHECK(t.stride(2) == 1 && t.stride(1) == t.size(2),
+              "head/dim axes must be dense");
+  return t.stride(0);
+}
 
 static hipStream_t cur_stream() {
   return (hipStream_t)c10::hip::getCurrentHIPStream().stream();
@@ -71,8 +81,11 @@ static void rope_cache(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        torch::Tensor positions, torch::Tensor slot_mapping,
                        torch::Tensor k_cache, torch::Tensor v_cache,
                        torch::Tensor cos_sin) {
-  CHECK_CUDA(q); CHECK_CONT(q); CHECK_CONT(k); CHECK_CONT(v);
+  CHECK_CUDA(q);
   CHECK_CONT(k_cache); CHECK_CONT(v_cache); CHECK_CONT(cos_sin);
+  const int64_t q_stride = q_token_stride(q);
+  const int64_t kv_stride = q_token_stride(k);
+  TORCH_CHECK(v.stride(0) == kv_stride, "k/v must share layout");
   TORCH_CHECK(positions.scalar_type() == at::kLong && slot_mapping.scalar_type() == at::kLong);
   TORCH_CHECK(cos_sin.scalar_type() == at::kFloat);
   const int N = q.size(0);
@@ -84,14 +97,15 @@ static void rope_cache(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   launch_rope_cache(q.data_ptr(), k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
                     v_cache.data_ptr(), positions.data_ptr<int64_t>(),
                     slot_mapping.data_ptr<int64_t>(), cos_sin.data_ptr<float>(), N,
-                    Hq, Hkv, D, P, cur_stream());
+                    Hq, Hkv, D, P, q_stride, kv_stride, cur_stream());
 }
 
 static void decode_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                         torch::Tensor v_cache, torch::Tensor block_tables,
                         torch::Tensor seq_lens, double scale) {
-  CHECK_CUDA(q); CHECK_CONT(q); CHECK_BF16(q); CHECK_CONT(out);
+  CHECK_CUDA(q); CHECK_BF16(q); CHECK_CONT(out);
   CHECK_CONT(k_cache); CHECK_CONT(v_cache);
+  const int64_t q_stride = q_token_stride(q);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt && seq_lens.scalar_type() == at::kInt);
   CHECK_CONT(block_tables); CHECK_CONT(seq_lens);
   const int B = q.size(0), Hq = q.size(1), D = q.size(2);
@@ -103,7 +117,7 @@ static void decode_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cach
   launch_decode_attn(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
                      v_cache.data_ptr(), block_tables.data_ptr<int>(),
                      seq_lens.data_ptr<int>(), (float)scale, B, Hq, Hkv, D,
-                     max_blocks, kv_block, cur_stream());
+                     max_blocks, kv_block, q_stride, cur_stream());
 }
 
 static void prefill_attn_impl(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
@@ -111,7 +125,8 @@ static void prefill_attn_impl(torch::Tensor out, torch::Tensor q, torch::Tensor 
                               torch::Tensor seq_lens, torch::Tensor ctx_lens,
                               torch::Tensor row_starts, torch::Tensor tile_seq,
                               torch::Tensor tile_q0, double scale, bool mfma) {
-  CHECK_CUDA(q); CHECK_CONT(q); CHECK_BF16(q); CHECK_CONT(out);
+  CHECK_CUDA(q); CHECK_BF16(q); CHECK_CONT(out);
+  const int64_t q_stride = q_token_stride(q);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt);
   const int Hq = q.size(1), D = q.size(2);
   const int Hkv = k_cache.size(2);
@@ -125,7 +140,7 @@ static void prefill_attn_impl(torch::Tensor out, torch::Tensor q, torch::Tensor 
      seq_lens.data_ptr<int>(), ctx_lens.data_ptr<int>(),
      row_starts.data_ptr<int>(), tile_seq.data_ptr<int>(),
      tile_q0.data_ptr<int>(), (float)scale, num_tiles, Hq, Hkv, D,
-     max_blocks, kv_block, cur_stream());
+     max_blocks, kv_block, q_stride, cur_stream());
 }
 
 static void prefill_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,

@@ -26,7 +26,7 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
     const int* __restrict__ block_tables,  // [B, max_blocks]
     const int* __restrict__ seq_lens,      // [B]
     const float scale, const int Hkv, const int max_blocks,
-    const int kv_block) {
+    const int kv_block, const int64_t q_stride) {
   const int seq = blockIdx.x;
   const int kvh = blockIdx.y;
   const int S = seq_lens[seq];
@@ -49,7 +49,7 @@ __global__ __launch_bounds__(256) void decode_attn_kernel(
 #pragma unroll
   for (int g = 0; g < G; ++g) {
     const int qh = kvh * G + g;
-    BF16x8 v8 = *(const BF16x8*)(q + ((int64_t)seq * Hq + qh) * D_HEAD + (sub << 3));
+    BF16x8 v8 = *(const BF16x8*)(q + (int64_t)seq * q_stride + (int64_t)qh * D_HEAD + (sub << 3));
 #pragma unroll
     for (int e = 0; e < 8; ++e) qreg[g][e] = bf2f(v8.h[e]) * scale;
   }
@@ -130,7 +130,8 @@ extern "C" void launch_decode_attn(void* out, const void* q,
                                    const int* block_tables,
                                    const int* seq_lens, float scale, int B,
                                    int Hq, int Hkv, int D, int max_blocks,
-                                   int kv_block, hipStream_t stream) {
+                                   int kv_block, int64_t q_stride,
+                                   hipStream_t stream) {
   if (D != D_HEAD) {
     // head_dim is 128 for every production preset; other sizes take the
     // python-assembled fallback path (ops/hip.py raises instead).
@@ -144,7 +145,7 @@ extern "C" void launch_decode_attn(void* out, const void* q,
                        (bf16_t*)out, (const bf16_t*)q,                    \
                        (const bf16_t*)k_cache, (const bf16_t*)v_cache,    \
                        block_tables, seq_lens, scale, Hkv, max_blocks,    \
-                       kv_block);                                         \
+                       kv_block, q_stride);                               \
     break;
   switch (G) {
     CASE(1) CASE(2) CASE(4) CASE(8)

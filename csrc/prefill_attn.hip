@@ -35,7 +35,7 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
     const int* __restrict__ tile_seq,      // [num_tiles] seq index per tile
     const int* __restrict__ tile_q0,       // [num_tiles] first row in chunk
     const float scale, const int Hq, const int Hkv, const int max_blocks,
-    const int kv_block) {
+    const int kv_block, const int64_t q_stride) {
   const int tile = blockIdx.x;
   const int head = blockIdx.y;
   const int kvh = head / (Hq / Hkv);
@@ -67,7 +67,8 @@ __global__ __launch_bounds__(256) void prefill_attn_kernel(
   const bool live = row < rows_here;
   const int qpos = ctx + q0 + row;        // global position of this q-row
   if (live) {
-    const bf16_t* qrow = q + ((int64_t)(row_base + q0 + row) * Hq + head) * D_HEAD + d0;
+    const bf16_t* qrow =
+        q + (int64_t)(row_base + q0 + row) * q_stride + (int64_t)head * D_HEAD + d0;
 #pragma unroll
     for (int e = 0; e < DIMS_PER_LANE; e += 8) {
       BF16x8 v8 = *(const BF16x8*)(qrow + e);
@@ -142,12 +143,12 @@ extern "C" void launch_prefill_attn(
     const int* block_tables, const int* seq_lens, const int* ctx_lens,
     const int* row_starts, const int* tile_seq, const int* tile_q0,
     float scale, int num_tiles, int Hq, int Hkv, int D, int max_blocks,
-    int kv_block, hipStream_t stream) {
+    int kv_block, int64_t q_stride, hipStream_t stream) {
   if (D != D_HEAD) return;
   dim3 grid(num_tiles, Hq), block(256);
   hipLaunchKernelGGL(prefill_attn_kernel, grid, block, 0, stream,
                      (bf16_t*)out, (const bf16_t*)q, (const bf16_t*)k_cache,
                      (const bf16_t*)v_cache, block_tables, seq_lens,
                      ctx_lens, row_starts, tile_seq, tile_q0, scale, Hq,
-                     Hkv, max_blocks, kv_block);
+                     Hkv, max_blocks, kv_block, q_stride);
 }

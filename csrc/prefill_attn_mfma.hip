@@ -39,7 +39,7 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
     const int* __restrict__ tile_seq,
     const int* __restrict__ tile_q0,
     const float scale, const int Hq, const int Hkv, const int max_blocks,
-    const int kv_block) {
+    const int kv_block, const int64_t q_stride) {
   const int tile = blockIdx.x;
   const int head = blockIdx.y;
   const int kvh = head / (Hq / Hkv);
@@ -74,7 +74,8 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
   const int my_qpos = q_live ? (ctx + my_qrow) : 0x3fffffff;  // OOB: never masked
   bf16x8_v qfrag[8];
   if (q_live) {
-    const bf16_t* qrow = q + ((int64_t)(row_base + my_qrow) * Hq + head) * D_HEAD;
+    const bf16_t* qrow =
+        q + (int64_t)(row_base + my_qrow) * q_stride + (int64_t)head * D_HEAD;
 #pragma unroll
     for (int kk = 0; kk < 8; ++kk)
       qfrag[kk] = *(const bf16x8_v*)(qrow + kk * 16 + khalf * 8);
@@ -234,12 +235,12 @@ extern "C" void launch_prefill_attn_mfma(
     const int* block_tables, const int* seq_lens, const int* ctx_lens,
     const int* row_starts, const int* tile_seq, const int* tile_q0,
     float scale, int num_tiles, int Hq, int Hkv, int D, int max_blocks,
-    int kv_block, hipStream_t stream) {
+    int kv_block, int64_t q_stride, hipStream_t stream) {
   if (D != D_HEAD) return;
   dim3 grid(num_tiles, Hq), block(256);
   hipLaunchKernelGGL(prefill_attn_mfma_kernel, grid, block, 0, stream,
                      (bf16_t*)out, (const bf16_t*)q, (const bf16_t*)k_cache,
                      (const bf16_t*)v_cache, block_tables, seq_lens,
                      ctx_lens, row_starts, tile_seq, tile_q0, scale, Hq,
-                     Hkv, max_blocks, kv_block);
+                     Hkv, max_blocks, kv_block, q_stride);
 }

@@ -12,15 +12,16 @@
 #include "common.h"
 
 __global__ void rope_cache_kernel(
-    bf16_t* __restrict__ q,            // [N, Hq, D]
-    bf16_t* __restrict__ k,            // [N, Hkv, D]
-    const bf16_t* __restrict__ v,      // [N, Hkv, D]
+    bf16_t* __restrict__ q,            // [N, Hq, D], token stride q_stride
+    bf16_t* __restrict__ k,            // [N, Hkv, D], token stride kv_stride
+    const bf16_t* __restrict__ v,      // [N, Hkv, D], token stride kv_stride
     bf16_t* __restrict__ k_cache,      // [slots, Hkv, D]
     bf16_t* __restrict__ v_cache,
     const int64_t* __restrict__ positions,     // [N]
     const int64_t* __restrict__ slot_mapping,  // [N]
     const float* __restrict__ cos_sin,         // [2, P, D/2]
-    const int Hq, const int Hkv, const int D, const int64_t P) {
+    const int Hq, const int Hkv, const int D, const int64_t P,
+    const int64_t q_stride, const int64_t kv_stride) {
   const int token = blockIdx.x;
   const int64_t pos = positions[token];
   const int64_t slot = slot_mapping[token];
@@ -35,7 +36,7 @@ __global__ void rope_cache_kernel(
 
   // Q heads: rope in place
   for (int hq = wid; hq < Hq; hq += nw) {
-    bf16_t* qh = q + ((int64_t)token * Hq + hq) * D;
+    bf16_t* qh = q + (int64_t)token * q_stride + (int64_t)hq * D;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {        // supports D up to 512
       if (r >= per_lane) break;
@@ -48,8 +49,8 @@ __global__ void rope_cache_kernel(
   }
   // K heads: rope in place + cache scatter; V heads: cache scatter
   for (int hk = wid; hk < Hkv; hk += nw) {
-    bf16_t* kh = k + ((int64_t)token * Hkv + hk) * D;
-    const bf16_t* vh = v + ((int64_t)token * Hkv + hk) * D;
+    bf16_t* kh = k + (int64_t)token * kv_stride + (int64_t)hk * D;
+    const bf16_t* vh = v + (int64_t)token * kv_stride + (int64_t)hk * D;
     bf16_t* kc = k_cache + (slot * Hkv + hk) * D;
     bf16_t* vc = v_cache + (slot * Hkv + hk) * D;
 #pragma unroll
@@ -75,10 +76,11 @@ extern "C" void launch_rope_cache(void* q, void* k, const void* v,
                                   const int64_t* slot_mapping,
                                   const float* cos_sin, int N, int Hq,
                                   int Hkv, int D, int64_t P,
+                                  int64_t q_stride, int64_t kv_stride,
                                   hipStream_t stream) {
   dim3 grid(N), block(256);
   hipLaunchKernelGGL(rope_cache_kernel, grid, block, 0, stream, (bf16_t*)q,
                      (bf16_t*)k, (const bf16_t*)v, (bf16_t*)k_cache,
                      (bf16_t*)v_cache, positions, slot_mapping, cos_sin, Hq,
-                     Hkv, D, P);
+                     Hkv, D, P, q_stride, kv_stride);
 }
