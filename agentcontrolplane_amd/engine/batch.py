@@ -72,6 +72,54 @@ class FlatBatch:
         return self._decode_lens_i32
 
 
+def batch_to_wire(batch: FlatBatch) -> dict:
+    """Compact picklable form for the TP metadata broadcast (rank 0 runs the
+    scheduler; other ranks rebuild the batch and run the sharded forward)."""
+    return {
+        "token_ids": batch.token_ids.tolist(),
+        "positions": batch.positions.tolist(),
+        "slot_mapping": batch.slot_mapping.tolist(),
+        "prefills": [
+            (m.seq_id, m.query_len, m.seq_len, m.ctx_len, m.block_table, m.needs_logits)
+            for m in batch.prefills
+        ],
+        "num_prefill_tokens": batch.num_prefill_tokens,
+        "decode_seq_ids": batch.decode_seq_ids,
+        "decode_block_tables": (
+            batch.decode_block_tables.tolist() if batch.decode_block_tables is not None else None
+        ),
+        "decode_seq_lens": (
+            batch.decode_seq_lens.tolist() if batch.decode_seq_lens is not None else None
+        ),
+        "logit_rows": batch.logit_rows.tolist(),
+        "sample_seq_ids": batch.sample_seq_ids,
+    }
+
+
+def batch_from_wire(d: dict, device) -> FlatBatch:
+    t = lambda x: torch.tensor(x, device=device, dtype=torch.long)  # noqa: E731
+    return FlatBatch(
+        token_ids=t(d["token_ids"]),
+        positions=t(d["positions"]),
+        slot_mapping=t(d["slot_mapping"]),
+        prefills=[
+            SeqMeta(seq_id=s, query_len=q, seq_len=sl, ctx_len=c, block_table=bt,
+                    needs_logits=nl)
+            for (s, q, sl, c, bt, nl) in d["prefills"]
+        ],
+        num_prefill_tokens=d["num_prefill_tokens"],
+        decode_seq_ids=d["decode_seq_ids"],
+        decode_block_tables=(
+            t(d["decode_block_tables"]) if d["decode_block_tables"] is not None else None
+        ),
+        decode_seq_lens=(
+            t(d["decode_seq_lens"]) if d["decode_seq_lens"] is not None else None
+        ),
+        logit_rows=t(d["logit_rows"]),
+        sample_seq_ids=d["sample_seq_ids"],
+    )
+
+
 @dataclasses.dataclass
 class PrefillMeta:
     """Device tensors for the batched prefill-attention kernel, built once

@@ -36,11 +36,27 @@ class InferenceEngine:
         self.mcfg = config.model_config()
         self.device = torch.device(config.device)
         self.tokenizer = ByteTokenizer(self.mcfg.vocab_size)
+        self.tp_world = config.tensor_parallel
+        self.tp_rank = 0
         if model is not None:
             self.model = model
         else:
-            self.model = create_model(self.mcfg, config, config.device)
-            self.model.random_init(config.seed)
+            if self.tp_world > 1:
+                import torch.distributed as dist
+
+                assert dist.is_initialized(), "tensor_parallel > 1 needs torch.distributed"
+                self.tp_rank = dist.get_rank()
+                from ..parallel.tp import make_all_reduce
+
+                self.model = create_model(
+                    self.mcfg, config, config.device,
+                    tp_rank=self.tp_rank, tp_world=self.tp_world,
+                )
+                self.model.random_init(config.seed)
+                self.model.all_reduce = make_all_reduce()
+            else:
+                self.model = create_model(self.mcfg, config, config.device)
+                self.model.random_init(config.seed)
         num_blocks = self._size_kv_pool()
         # +1 scratch block: hipGraph decode padding rows write their KV there
         self.model.allocate_kv_cache(num_blocks + 1, config.kv_block_size)
@@ -117,6 +133,13 @@ class InferenceEngine:
         if self._thread is not None:
             self._thread.join(timeout=10)
             self._thread = None
+        if self.tp_world > 1 and self.tp_rank == 0:
+            try:
+                from ..parallel.tp import STOP, broadcast_step
+
+                broadcast_step(STOP)
+            except Exception:
+                pass
 
     # ------------------------------------------------------------- submit
 
@@ -256,6 +279,11 @@ class InferenceEngine:
         if out.batch is None:
             return False
         batch = out.batch
+        if self.tp_world > 1 and self.tp_rank == 0:
+            from ..engine.batch import batch_to_wire
+            from ..parallel.tp import broadcast_step
+
+            broadcast_step(batch_to_wire(batch))
         logits = None
         if self.graph_runner is not None and not batch.prefills:
             logits = self.graph_runner.run(batch)

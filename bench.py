@@ -167,11 +167,17 @@ def main() -> None:
         request_timeout_s=1200,
     )
     if args.tp > 1:
-        raise NotImplementedError(
-            "tensor-parallel serving runs through parallel/tp.py's sharded "
-            "engine worker; the driver bench uses DP request sharding"
-        )
-    engine = InferenceEngine(ecfg)
+        # TP serving (config 3): rank 0 runs the control plane + scheduler;
+        # ranks 1..tp-1 follow the metadata broadcast (parallel/tp.py)
+        assert world == args.tp, "launch with torchrun --nproc-per-node == --tp"
+        engine = InferenceEngine(ecfg, start=(rank == 0))
+        if rank != 0:
+            from agentcontrolplane_amd.parallel.tp import run_tp_worker
+
+            run_tp_worker(engine)
+            return
+    else:
+        engine = InferenceEngine(ecfg)
 
     cp = ControlPlane(engine=engine, auto_approve="approve", llm_probe=False)
     cp.start()
@@ -222,7 +228,8 @@ def main() -> None:
         for _ in range(args.warmup):
             run_wave(cp, "bench-agent", conc, rng)
 
-        if is_dist:
+        dp_sync = is_dist and args.tp <= 1  # TP workers sit in their own loop
+        if dp_sync:
             dist.barrier()
         if device == "cuda":
             torch.cuda.synchronize()
@@ -232,16 +239,16 @@ def main() -> None:
             all_lat.extend(run_wave(cp, "bench-agent", conc, rng))
         if device == "cuda":
             torch.cuda.synchronize()
-        if is_dist:
+        if dp_sync:
             dist.barrier()
         elapsed = time.monotonic() - t_start
-        if is_dist:
+        if dp_sync:
             t = torch.tensor([elapsed])
             dist.all_reduce(t, op=dist.ReduceOp.MAX)
             elapsed = float(t[0])
 
         agent_steps = 2 * conc * args.steps  # 2 LLM turns per task per wave
-        total_steps = agent_steps * world
+        total_steps = agent_steps * (1 if args.tp > 1 else world)
         value = total_steps / elapsed
         p50 = statistics.median(all_lat)
         em = engine.metrics()
@@ -261,7 +268,7 @@ def main() -> None:
                 "data": "synthetic (random-init weights, pseudo-random prompts; no network)",
                 "config": {
                     "model": model,
-                    "global_batch": conc * world,
+                    "global_batch": conc * (1 if args.tp > 1 else world),
                     "seq_len": 8192,
                     "parallelism": (f"tp{args.tp}" if args.tp > 1 else f"dp{world}"),
                     "concurrency_per_gpu": conc,
