@@ -112,6 +112,10 @@ class MixtralForCausalLM(LlamaForCausalLM):
                 o = self.all_reduce(o)
             normed2, residual = ops.fused_add_rmsnorm(o, residual, lw.post_norm, cfg.rms_eps)
             x = self._moe_mlp(li, normed2)
+            if self.all_reduce is not None and self.moe_dispatch is None:
+                # TP-sharded experts (intermediate split): partial sums reduce
+                # here; EP dispatch returns fully-combined rows instead
+                x = self.all_reduce(x)
         normed, _ = ops.fused_add_rmsnorm(x, residual, self.final_norm, cfg.rms_eps)
         sel = normed[batch.logit_rows]
         return F.linear(sel, self.lm_head).float()
