@@ -1,0 +1,131 @@
+"""CLI — the cmd/main.go equivalent.
+
+    python -m agentcontrolplane_amd serve [--wal PATH] [--model llama3-8b]
+        [--device cuda|cpu|none] [--port 8082] [--auto-approve]
+    python -m agentcontrolplane_amd apply -f manifest.yaml [-f more.yaml]
+        [--server http://127.0.0.1:8082]
+    python -m agentcontrolplane_amd get <kind> [name]
+
+``serve`` wires the store + six reconcilers + MCP manager + optional local
+engine + the REST API on :8082 (the reference's api_service port).
+``apply`` accepts the same YAML manifests the reference's CRDs use
+(acp.humanlayer.dev/v1alpha1 Agent/LLM/Task/…), posted into a running
+server's store through a small admin endpoint, so existing manifests run
+unchanged without a kubernetes cluster.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+
+
+def cmd_serve(args) -> None:
+    import uvicorn
+
+    from .engine.config import EngineConfig
+    from .runtime import ControlPlane
+
+    engine = None
+    if args.device != "none":
+        import torch
+
+        device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+        model = args.model if device == "cuda" else "tiny"
+        from .engine.engine import InferenceEngine
+
+        engine = InferenceEngine(
+            EngineConfig(
+                model=model,
+                device=device,
+                checkpoint_path=args.checkpoint,
+                num_kv_blocks=args.kv_blocks,
+            )
+        )
+    cp = ControlPlane(
+        wal_path=args.wal,
+        engine=engine,
+        auto_approve="approve" if args.auto_approve else None,
+    )
+    cp.start()
+    app = cp.rest_app
+    from .server.admin import add_admin_routes
+
+    add_admin_routes(app, cp.store)
+    print(f"acp-amd serving on :{args.port} (engine={'on' if engine else 'off'})")
+    try:
+        uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+    finally:
+        cp.stop()
+        if engine is not None:
+            engine.stop()
+
+
+def cmd_apply(args) -> None:
+    import yaml
+
+    try:
+        import httpx
+
+        client = httpx.Client(base_url=args.server, timeout=30)
+    except ImportError:  # pragma: no cover
+        print("httpx required for apply", file=sys.stderr)
+        sys.exit(1)
+    for path in args.files:
+        with open(path) as f:
+            for doc in yaml.safe_load_all(f):
+                if not doc:
+                    continue
+                r = client.post("/admin/resources", json=doc)
+                name = doc.get("metadata", {}).get("name", "?")
+                kind = doc.get("kind", "?")
+                if r.status_code in (200, 201):
+                    print(f"{kind.lower()}/{name} {'configured' if r.status_code == 200 else 'created'}")
+                else:
+                    print(f"{kind.lower()}/{name} error: {r.text}", file=sys.stderr)
+
+
+def cmd_get(args) -> None:
+    import httpx
+
+    client = httpx.Client(base_url=args.server, timeout=30)
+    url = f"/admin/resources/{args.kind}"
+    if args.name:
+        url += f"/{args.name}"
+    r = client.get(url, params={"namespace": args.namespace})
+    print(json.dumps(r.json(), indent=2))
+
+
+def main() -> None:
+    p = argparse.ArgumentParser(prog="agentcontrolplane_amd")
+    sub = p.add_subparsers(dest="cmd", required=True)
+
+    s = sub.add_parser("serve", help="run the control plane + engine + REST API")
+    s.add_argument("--host", default="127.0.0.1")
+    s.add_argument("--port", type=int, default=8082)
+    s.add_argument("--wal", default=None, help="WAL path for durable state")
+    s.add_argument("--model", default="llama3-8b")
+    s.add_argument("--checkpoint", default=None, help="safetensors checkpoint dir")
+    s.add_argument("--device", default=None, help="cuda | cpu | none (no engine)")
+    s.add_argument("--kv-blocks", type=int, default=None)
+    s.add_argument("--auto-approve", action="store_true")
+    s.set_defaults(fn=cmd_serve)
+
+    a = sub.add_parser("apply", help="apply CRD-style YAML manifests")
+    a.add_argument("-f", "--files", action="append", required=True)
+    a.add_argument("--server", default="http://127.0.0.1:8082")
+    a.set_defaults(fn=cmd_apply)
+
+    g = sub.add_parser("get", help="get resources")
+    g.add_argument("kind")
+    g.add_argument("name", nargs="?")
+    g.add_argument("--namespace", default="default")
+    g.add_argument("--server", default="http://127.0.0.1:8082")
+    g.set_defaults(fn=cmd_get)
+
+    args = p.parse_args()
+    args.fn(args)
+
+
+if __name__ == "__main__":
+    main()

@@ -129,6 +129,8 @@ class EngineConfig:
     tensor_parallel: int = 1
     seed: int = 0
     request_timeout_s: float = 600.0
+    # HF-named safetensors dir; None = random-init (no network here)
+    checkpoint_path: Optional[str] = None
 
     def model_config(self) -> ModelConfig:
         if self.model not in PRESETS:

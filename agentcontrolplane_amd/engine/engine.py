@@ -52,11 +52,11 @@ class InferenceEngine:
                     self.mcfg, config, config.device,
                     tp_rank=self.tp_rank, tp_world=self.tp_world,
                 )
-                self.model.random_init(config.seed)
+                self._init_weights()
                 self.model.all_reduce = make_all_reduce()
             else:
                 self.model = create_model(self.mcfg, config, config.device)
-                self.model.random_init(config.seed)
+                self._init_weights()
         num_blocks = self._size_kv_pool()
         # +1 scratch block: hipGraph decode padding rows write their KV there
         self.model.allocate_kv_cache(num_blocks + 1, config.kv_block_size)
@@ -101,6 +101,14 @@ class InferenceEngine:
         self._start_time = time.monotonic()
         if start:
             self.start()
+
+    def _init_weights(self) -> None:
+        if self.cfg.checkpoint_path:
+            from ..models.weights import load_checkpoint
+
+            load_checkpoint(self.model, self.cfg.checkpoint_path)
+        else:
+            self.model.random_init(self.cfg.seed)
 
     # ------------------------------------------------------------- sizing
 

@@ -106,16 +106,12 @@ class LlamaForCausalLM:
             lw.down = randw(h, self.inter)
             self.layers.append(lw)
 
-    def load_safetensors(self, path: str) -> None:  # pragma: no cover - no checkpoints here
-        """Checkpoint loading seam (safetensors shard dir, HF llama naming).
-        Unused in this deployment (no network for weights) but kept so a real
-        checkpoint drops in."""
-        import json
-        import os
+    def load_safetensors(self, path: str) -> None:
+        """Load an HF-named safetensors checkpoint dir (round-trip tested in
+        tests/test_checkpoint.py; handles TP slicing)."""
+        from .weights import load_checkpoint
 
-        from safetensors.torch import load_file
-
-        raise NotImplementedError("checkpoint loading lands with a reachable weights source")
+        load_checkpoint(self, path)
 
     # ------------------------------------------------------------ KV cache
 
