@@ -133,6 +133,11 @@ def main() -> None:
     p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree (else DP sharding)")
     p.add_argument("--device", default=None)
     p.add_argument("--kv-blocks", type=int, default=None)
+    p.add_argument(
+        "--no-wal", action="store_true",
+        help="disable the durable WAL (default on: the reference checkpoints "
+        "every transition to etcd, so the measured loop includes durability)",
+    )
     args = p.parse_args()
 
     import torch
@@ -179,7 +184,16 @@ def main() -> None:
     else:
         engine = InferenceEngine(ecfg)
 
-    cp = ControlPlane(engine=engine, auto_approve="approve", llm_probe=False)
+    wal_path = None
+    if not args.no_wal:
+        import tempfile
+
+        wal_path = os.path.join(
+            tempfile.mkdtemp(prefix="acp-bench-"), f"wal-rank{rank}.jsonl"
+        )
+    cp = ControlPlane(
+        engine=engine, auto_approve="approve", llm_probe=False, wal_path=wal_path
+    )
     cp.start()
     try:
         cp.store.create(
