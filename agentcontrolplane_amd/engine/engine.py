@@ -81,6 +81,8 @@ class InferenceEngine:
             )
         self._gen = torch.Generator(device=self.device.type)
         self._gen.manual_seed(config.seed)
+        self._sampling_cache_key = None
+        self._sampling_cache = None
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
         self._pending: List[InferenceRequest] = []
@@ -317,26 +319,39 @@ class InferenceEngine:
         if B == 0:
             return
         live = logits[:, :N_SPECIAL]  # sampling restricted to decodable ids
-        temps = torch.tensor(
-            [s.request.sampling.temperature for s in seqs], device=logits.device
-        )
-        top_ks = torch.tensor(
-            [s.request.sampling.top_k for s in seqs], device=logits.device, dtype=torch.long
-        )
-        top_ps = torch.tensor([s.request.sampling.top_p for s in seqs], device=logits.device)
+        # per-sequence sampling params are static: cache the device tensors
+        # against the batch's id tuple (stable across decode steps)
+        key = tuple(batch.sample_seq_ids)
+        cached = self._sampling_cache if self._sampling_cache_key == key else None
+        if cached is None:
+            temps = torch.tensor(
+                [s.request.sampling.temperature for s in seqs], device=logits.device
+            )
+            top_ks = torch.tensor(
+                [s.request.sampling.top_k for s in seqs], device=logits.device,
+                dtype=torch.long,
+            )
+            top_ps = torch.tensor(
+                [s.request.sampling.top_p for s in seqs], device=logits.device
+            )
+            self._sampling_cache_key = key
+            self._sampling_cache = (temps, top_ks, top_ps)
+        else:
+            temps, top_ks, top_ps = cached
         mask = None
         any_grammar = any(s.grammar is not None for s in seqs)
         if any_grammar:
-            mask = torch.ones(B, N_SPECIAL, dtype=torch.bool)
+            import numpy as np
+
+            m = np.ones((B, N_SPECIAL), dtype=bool)
             for i, s in enumerate(seqs):
                 if s.grammar is not None:
                     allowed = s.grammar.allowed_tokens()
                     if s.grammar.accepting:
                         allowed = set(allowed) | {EOT}
-                    row = torch.zeros(N_SPECIAL, dtype=torch.bool)
-                    row[list(allowed)] = True
-                    mask[i] = row
-            mask = mask.to(logits.device)
+                    m[i] = False
+                    m[i, list(allowed)] = True
+            mask = torch.from_numpy(m).to(logits.device, non_blocking=True)
         tokens = ops.softmax_sample(live, temps, top_ks, top_ps, self._gen, mask)
         tokens_cpu = tokens.tolist()
         for s, tok in zip(seqs, tokens_cpu):

@@ -29,6 +29,9 @@ class PyBlockManager:
         self._free: List[int] = list(range(num_blocks - 1, -1, -1))
         self._tables: Dict[int, List[int]] = {}
         self._lens: Dict[int, int] = {}
+        #: bumps whenever any block table changes shape — callers cache
+        #: derived tensors against it
+        self.table_epoch = 0
 
     # ------------------------------------------------------------ lifecycle
 
@@ -43,6 +46,7 @@ class PyBlockManager:
         self._lens.pop(seq_id, None)
         if blocks:
             self._free.extend(reversed(blocks))
+            self.table_epoch += 1
 
     def has_seq(self, seq_id: int) -> bool:
         return seq_id in self._tables
@@ -68,6 +72,8 @@ class PyBlockManager:
                 f"seq {seq_id}: need {need} blocks, {len(self._free)} free"
             )
         table = self._tables[seq_id]
+        if need:
+            self.table_epoch += 1
         for _ in range(need):
             table.append(self._free.pop())
         slots = []

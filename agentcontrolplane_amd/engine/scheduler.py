@@ -70,13 +70,19 @@ class Scheduler:
         self.device = torch.device(device)
         self.waiting: List[Sequence] = []
         self.running: List[Sequence] = []   # in admission order (oldest first)
+        self._by_id: Dict[int, Sequence] = {}
         self._seq_counter = 0
+        # decode block-table tensor cache: rebuilt only when membership or
+        # any member's block list changes (the table epoch)
+        self._decode_cache_key = None
+        self._decode_cache = None
 
     def add_request(self, request: InferenceRequest) -> Sequence:
         self._seq_counter += 1
         seq = Sequence(self._seq_counter, request)
         request.seq = seq
         self.waiting.append(seq)
+        self._by_id[seq.seq_id] = seq
         return seq
 
     def has_work(self) -> bool:
@@ -197,10 +203,9 @@ class Scheduler:
 
         decode_block_tables = None
         decode_seq_lens = None
+        decode_tables_i32 = None
         decode_ids: List[int] = []
         if decode_seqs:
-            max_blocks = 0
-            tables = []
             lens = []
             for s in decode_seqs:
                 tok = s.output_ids[-1] if s.output_ids else s.prompt_ids[-1]
@@ -209,16 +214,26 @@ class Scheduler:
                 token_ids.append(tok)
                 positions.append(pos)
                 slot_mapping.extend(slots)
-                t = self.bm.block_table(s.seq_id)
-                tables.append(list(t))
                 lens.append(pos + 1)
-                max_blocks = max(max_blocks, len(t))
                 logit_rows.append(row)
                 sample_seq_ids.append(s.seq_id)
                 decode_ids.append(s.seq_id)
                 row += 1
-            padded = [t + [0] * (max_blocks - len(t)) for t in tables]
-            decode_block_tables = torch.tensor(padded, device=self.device, dtype=torch.long)
+            # block tables change only when membership changes or a table
+            # grows (one block per kv_block_size steps): cache the tensor
+            key = (tuple(decode_ids), self.bm.table_epoch)
+            if key == self._decode_cache_key:
+                decode_block_tables, decode_tables_i32 = self._decode_cache
+            else:
+                tables = [self.bm.block_table(i) for i in decode_ids]
+                max_blocks = max(len(t) for t in tables)
+                padded = [t + [0] * (max_blocks - len(t)) for t in tables]
+                decode_block_tables = torch.tensor(
+                    padded, device=self.device, dtype=torch.long
+                )
+                decode_tables_i32 = decode_block_tables.int()
+                self._decode_cache_key = key
+                self._decode_cache = (decode_block_tables, decode_tables_i32)
             decode_seq_lens = torch.tensor(lens, device=self.device, dtype=torch.long)
 
         batch = FlatBatch(
@@ -233,15 +248,14 @@ class Scheduler:
             logit_rows=torch.tensor(logit_rows, device=self.device, dtype=torch.long),
             sample_seq_ids=sample_seq_ids,
         )
+        if decode_tables_i32 is not None:
+            batch._decode_tables_i32 = decode_tables_i32
         return SchedulerOutput(batch=batch, preempted=preempted)
 
     # ------------------------------------------------------------ commit
 
     def seq_by_id(self, seq_id: int) -> Optional[Sequence]:
-        for s in self.running:
-            if s.seq_id == seq_id:
-                return s
-        return None
+        return self._by_id.get(seq_id)
 
     def append_sampled(self, seq: Sequence, token: int) -> None:
         seq.output_ids.append(token)
@@ -253,6 +267,7 @@ class Scheduler:
             self.running.remove(seq)
         if self.bm.has_seq(seq.seq_id):
             self.bm.free_seq(seq.seq_id)
+        self._by_id.pop(seq.seq_id, None)
         seq.request.finish(reason)
 
     def _preempt(self, seq: Sequence) -> None:
@@ -275,4 +290,5 @@ class Scheduler:
             self.waiting.remove(seq)
         if self.bm.has_seq(seq.seq_id):
             self.bm.free_seq(seq.seq_id)
+        self._by_id.pop(seq.seq_id, None)
         seq.request.fail(err)

@@ -32,6 +32,7 @@ class BlockManager {
   void free_seq(int64_t seq_id) {
     auto it = tables_.find(seq_id);
     if (it == tables_.end()) return;
+    if (!it->second.empty()) ++table_epoch_;
     for (auto rit = it->second.rbegin(); rit != it->second.rend(); ++rit)
       free_.push_back(*rit);
     tables_.erase(it);
@@ -51,6 +52,7 @@ class BlockManager {
                                std::to_string(need) + " blocks, " +
                                std::to_string(free_.size()) + " free");
     auto& table = tables_.at(seq_id);
+    if (need > 0) ++table_epoch_;
     for (int64_t i = 0; i < need; ++i) {
       table.push_back(free_.back());
       free_.pop_back();
@@ -73,6 +75,7 @@ class BlockManager {
   int64_t seq_len(int64_t seq_id) const { return lens_.at(seq_id); }
 
   int64_t free_blocks() const { return (int64_t)free_.size(); }
+  int64_t table_epoch() const { return table_epoch_; }
   int64_t used_blocks() const { return num_blocks_ - (int64_t)free_.size(); }
   int64_t num_blocks() const { return num_blocks_; }
   int64_t block_size() const { return block_size_; }
@@ -89,6 +92,7 @@ class BlockManager {
   }
 
   int64_t num_blocks_, block_size_;
+  int64_t table_epoch_ = 0;
   std::vector<int64_t> free_;
   std::unordered_map<int64_t, std::vector<int64_t>> tables_;
   std::unordered_map<int64_t, int64_t> lens_;
@@ -107,6 +111,7 @@ void register_block_manager(py::module_& m) {
       .def("seq_len", &BlockManager::seq_len)
       .def("occupancy", &BlockManager::occupancy)
       .def_property_readonly("free_blocks", &BlockManager::free_blocks)
+      .def_property_readonly("table_epoch", &BlockManager::table_epoch)
       .def_property_readonly("used_blocks", &BlockManager::used_blocks)
       .def_property_readonly("num_blocks", &BlockManager::num_blocks)
       .def_property_readonly("block_size", &BlockManager::block_size);
