@@ -415,4 +415,7 @@ class InferenceEngine:
         m["kv_occupancy"] = self.bm.used_blocks / max(1, self.bm.num_blocks)
         m["running_seqs"] = len(self.scheduler.running)
         m["waiting_seqs"] = len(self.scheduler.waiting)
+        m["retired_seqs"] = len(self.scheduler.retired)
+        m["continuation_hits"] = self.scheduler.continuation_hits
+        m["continuation_tokens_saved"] = self.scheduler.continuation_tokens_saved
         return m

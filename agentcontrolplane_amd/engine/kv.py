@@ -84,6 +84,24 @@ class PyBlockManager:
         self._lens[seq_id] = start + n_tokens
         return slots
 
+    def adopt_prefix(self, new_seq_id: int, old_seq_id: int, n_blocks: int,
+                     n_tokens: int) -> None:
+        """Transfer the first n_blocks of old's table to a NEW sequence (the
+        continuation cache: a finished conversation's KV prefix is reused by
+        the turn that extends it); the remainder of old is freed."""
+        if new_seq_id in self._tables:
+            raise KeyError(f"seq {new_seq_id} already exists")
+        old = self._tables.pop(old_seq_id, None)
+        self._lens.pop(old_seq_id, None)
+        if old is None:
+            raise KeyError(f"seq {old_seq_id} not found")
+        assert n_blocks <= len(old) and n_tokens <= n_blocks * self.block_size
+        self._tables[new_seq_id] = old[:n_blocks]
+        self._lens[new_seq_id] = n_tokens
+        if len(old) > n_blocks:
+            self._free.extend(reversed(old[n_blocks:]))
+        self.table_epoch += 1
+
     # -------------------------------------------------------------- queries
 
     def block_table(self, seq_id: int) -> List[int]:

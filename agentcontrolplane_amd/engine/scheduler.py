@@ -76,6 +76,17 @@ class Scheduler:
         # any member's block list changes (the table epoch)
         self._decode_cache_key = None
         self._decode_cache = None
+        # continuation cache: finished sequences keep their KV blocks for a
+        # while; a new request whose prompt extends a finished conversation
+        # adopts the shared prefix instead of re-prefilling it (the agent
+        # loop's next turn always extends the previous one).  KV stays a
+        # derived cache — eviction only costs recompute.
+        self.retired: Dict[int, tuple] = {}      # seq_id -> token tuple (FIFO)
+        self._retired_index: Dict[tuple, List[int]] = {}
+        self.max_retired = 4096
+        self.continuation_cache = True
+        self.continuation_hits = 0
+        self.continuation_tokens_saved = 0
 
     def add_request(self, request: InferenceRequest) -> Sequence:
         self._seq_counter += 1
@@ -112,6 +123,8 @@ class Scheduler:
             decode_need = sum(seq_blocks_for(s, 1) for s in decode_seqs)
             if decode_need <= self.bm.free_blocks:
                 break
+            if self._evict_one_retired():
+                continue  # reclaimed cache blocks before preempting live work
             victim = decode_seqs.pop()  # youngest
             self.running.remove(victim)
             self._preempt(victim)
@@ -151,17 +164,33 @@ class Scheduler:
                     ),
                 )
                 continue
-            want = min(len(s.prompt_ids), budget)
-            chunk = fit_chunk(0, 0, want)
-            if chunk <= 0:
+            want0 = min(len(s.prompt_ids), budget)
+            if avail * bs < want0 and self.retired:
+                before = self.bm.free_blocks
+                self._reclaim(self.bm.free_blocks + (want0 + bs - 1) // bs)
+                avail += self.bm.free_blocks - before
+            if fit_chunk(0, 0, 1) <= 0:
                 break
             self.waiting.pop(0)
-            self.bm.add_seq(s.seq_id)
+            adopted = self._try_adopt(s)
+            if not adopted:
+                self.bm.add_seq(s.seq_id)
+                s.num_processed = 0
+            cur = self.bm.seq_len(s.seq_id)
+            have = len(self.bm.block_table(s.seq_id))
+            want = min(len(s.prompt_ids) - s.num_processed, budget)
+            chunk = fit_chunk(cur, have, want)
+            if chunk <= 0:
+                # adopted blocks but no room to extend this step: keep it
+                # running as a pending prefill for the next step
+                s.state = PREFILL
+                self.running.append(s)
+                continue
             s.state = PREFILL
             self.running.append(s)
             prefills.append((s, chunk))
             budget -= chunk
-            avail -= blocks_for(0, 0, chunk)
+            avail -= blocks_for(cur, have, chunk)
 
         decode_seqs = [s for s in self.running if s.state == DECODE]
         if not prefills and not decode_seqs:
@@ -266,9 +295,77 @@ class Scheduler:
         if seq in self.running:
             self.running.remove(seq)
         if self.bm.has_seq(seq.seq_id):
-            self.bm.free_seq(seq.seq_id)
+            if self.continuation_cache and self.bm.seq_len(seq.seq_id) >= self.cfg.kv_block_size:
+                self._retire(seq)
+            else:
+                self.bm.free_seq(seq.seq_id)
         self._by_id.pop(seq.seq_id, None)
         seq.request.finish(reason)
+
+    # ------------------------------------------------------- continuation
+
+    def _first_block_key(self, tokens) -> tuple:
+        return tuple(tokens[: self.cfg.kv_block_size])
+
+    def _retire(self, seq: Sequence) -> None:
+        tokens = tuple(seq.prompt_ids + seq.output_ids)
+        cached = self.bm.seq_len(seq.seq_id)
+        tokens = tokens[:cached]
+        self.retired[seq.seq_id] = tokens
+        self._retired_index.setdefault(self._first_block_key(tokens), []).append(seq.seq_id)
+        while len(self.retired) > self.max_retired:
+            self._evict_one_retired()
+
+    def _evict_one_retired(self) -> bool:
+        if not self.retired:
+            return False
+        old_id = next(iter(self.retired))
+        tokens = self.retired.pop(old_id)
+        bucket = self._retired_index.get(self._first_block_key(tokens))
+        if bucket is not None:
+            try:
+                bucket.remove(old_id)
+            except ValueError:
+                pass
+            if not bucket:
+                self._retired_index.pop(self._first_block_key(tokens), None)
+        self.bm.free_seq(old_id)
+        return True
+
+    def _reclaim(self, need_blocks: int) -> None:
+        while self.bm.free_blocks < need_blocks and self._evict_one_retired():
+            pass
+
+    def _try_adopt(self, seq: Sequence) -> bool:
+        """Adopt a retired conversation's KV prefix; returns True if the
+        sequence was registered with the block manager."""
+        bs = self.cfg.kv_block_size
+        if not self.continuation_cache or len(seq.prompt_ids) <= bs:
+            return False
+        key = self._first_block_key(seq.prompt_ids)
+        best_id, best_len = None, 0
+        for cand in self._retired_index.get(key, []):
+            toks = self.retired[cand]
+            limit = min(len(toks), len(seq.prompt_ids) - 1)
+            n = bs  # first block already known equal
+            while n < limit and toks[n] == seq.prompt_ids[n]:
+                n += 1
+            if n > best_len:
+                best_id, best_len = cand, n
+        n_blocks = (best_len // bs) if best_id is not None else 0
+        if n_blocks < 1:
+            return False
+        toks = self.retired.pop(best_id)
+        bucket = self._retired_index.get(key)
+        if bucket is not None:
+            bucket.remove(best_id)
+            if not bucket:
+                self._retired_index.pop(key, None)
+        self.bm.adopt_prefix(seq.seq_id, best_id, n_blocks, n_blocks * bs)
+        seq.num_processed = n_blocks * bs
+        self.continuation_hits += 1
+        self.continuation_tokens_saved += n_blocks * bs
+        return True
 
     def _preempt(self, seq: Sequence) -> None:
         """Recompute-style preemption: blocks freed, prompt grows to include

@@ -68,6 +68,26 @@ class BlockManager {
     return slots;
   }
 
+  void adopt_prefix(int64_t new_seq_id, int64_t old_seq_id, int64_t n_blocks,
+                    int64_t n_tokens) {
+    if (tables_.count(new_seq_id))
+      throw py::key_error("seq " + std::to_string(new_seq_id) + " already exists");
+    auto it = tables_.find(old_seq_id);
+    if (it == tables_.end())
+      throw py::key_error("seq " + std::to_string(old_seq_id) + " not found");
+    auto old = std::move(it->second);
+    tables_.erase(it);
+    lens_.erase(old_seq_id);
+    if (n_blocks > (int64_t)old.size() || n_tokens > n_blocks * block_size_)
+      throw std::runtime_error("adopt_prefix: bad prefix bounds");
+    tables_[new_seq_id] =
+        std::vector<int64_t>(old.begin(), old.begin() + n_blocks);
+    lens_[new_seq_id] = n_tokens;
+    for (auto rit = old.rbegin(); rit != old.rend() - n_blocks; ++rit)
+      free_.push_back(*rit);
+    ++table_epoch_;
+  }
+
   std::vector<int64_t> block_table(int64_t seq_id) const {
     return tables_.at(seq_id);
   }
@@ -107,6 +127,7 @@ void register_block_manager(py::module_& m) {
       .def("has_seq", &BlockManager::has_seq)
       .def("can_append", &BlockManager::can_append)
       .def("append_tokens", &BlockManager::append_tokens)
+      .def("adopt_prefix", &BlockManager::adopt_prefix)
       .def("block_table", &BlockManager::block_table)
       .def("seq_len", &BlockManager::seq_len)
       .def("occupancy", &BlockManager::occupancy)
