@@ -38,8 +38,12 @@ class ControlPlane:
         llm_probe: bool = True,
         pod_name: str = "acp-controller-0",
         fsync: str = "interval",
+        store: Optional[ResourceStore] = None,
     ):
-        self.store = ResourceStore(wal_path=wal_path, fsync=fsync)
+        # an injected store lets several ControlPlane replicas share one
+        # backing store (the reference's multi-pod deployment over one etcd)
+        self.store = store if store is not None else ResourceStore(wal_path=wal_path, fsync=fsync)
+        self._owns_store = store is None
         self.tracer = get_tracer()
         self.engine = engine
         self.mcp = MCPServerManager(self.store)
@@ -93,7 +97,8 @@ class ControlPlane:
     def stop(self) -> None:
         self.manager.stop()
         self.mcp.close()
-        self.store.close()
+        if self._owns_store:
+            self.store.close()
 
     def __enter__(self) -> "ControlPlane":
         return self.start()
