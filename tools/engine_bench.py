@@ -19,6 +19,7 @@ def main():
     p.add_argument("--decode-tokens", type=int, default=64)
     p.add_argument("--constrained", action="store_true")
     p.add_argument("--kv-blocks", type=int, default=None)
+    p.add_argument("--prefill-budget", type=int, default=8192)
     p.add_argument("--device", default=None)
     args = p.parse_args()
 
@@ -35,7 +36,7 @@ def main():
             model=model,
             device=device,
             num_kv_blocks=args.kv_blocks,
-            max_prefill_tokens=8192,
+            max_prefill_tokens=args.prefill_budget,
             request_timeout_s=1200,
         )
     )
