@@ -116,13 +116,15 @@ class LlamaForCausalLM:
     # ------------------------------------------------------------ KV cache
 
     def allocate_kv_cache(self, num_blocks: int, block_size: int) -> None:
+        # uninitialized on purpose: every slot is written (rope+scatter)
+        # before any attention reads it; zero-filling 200+ GB costs ~35 ms/GB
         shape = (num_blocks, block_size, self.n_kv_heads, self.head_dim)
         self.k_caches = [
-            torch.zeros(shape, dtype=self.dtype, device=self.device)
+            torch.empty(shape, dtype=self.dtype, device=self.device)
             for _ in range(self.cfg.num_layers)
         ]
         self.v_caches = [
-            torch.zeros(shape, dtype=self.dtype, device=self.device)
+            torch.empty(shape, dtype=self.dtype, device=self.device)
             for _ in range(self.cfg.num_layers)
         ]
 
@@ -135,21 +137,19 @@ class LlamaForCausalLM:
     # -------------------------------------------------------------- forward
 
     def _attention(self, layer_idx: int, q, batch: FlatBatch):
-        """q: [N, Hq, D] post-rope; returns [N, Hq, D]."""
+        """q: [N, Hq, D] post-rope; returns [N, Hq, D] (one preallocated
+        output; prefill and decode kernels write disjoint row ranges)."""
         kc, vc = self.k_caches[layer_idx], self.v_caches[layer_idx]
-        outs = []
+        N = q.shape[0]
+        out = torch.empty(
+            (N, self.n_heads, self.head_dim), dtype=self.dtype, device=q.device
+        )
         row = batch.num_prefill_tokens
         if batch.prefills:
-            outs.append(
-                ops.attention_prefill_batch(q[:row], kc, vc, batch, self.scale)
-            )
+            ops.attention_prefill_batch(q[:row], kc, vc, batch, self.scale, out=out[:row])
         if batch.num_decode:
-            outs.append(
-                ops.attention_decode_batch(
-                    q[row:], kc, vc, batch, self.scale,
-                )
-            )
-        return torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
+            ops.attention_decode_batch(q[row:], kc, vc, batch, self.scale, out=out[row:])
+        return out
 
     def forward(self, batch: FlatBatch) -> torch.Tensor:
         """Returns logits [len(batch.sample_seq_ids), vocab]."""

@@ -64,15 +64,15 @@ def attention_prefill(q, k_cache, v_cache, block_table, seq_len, ctx_len, scale)
     return _impl(q).attention_prefill(q, k_cache, v_cache, block_table, seq_len, ctx_len, scale)
 
 
-def attention_prefill_batch(q, k_cache, v_cache, batch, scale):
+def attention_prefill_batch(q, k_cache, v_cache, batch, scale, out=None):
     """Causal attention for ALL prefill chunks of a FlatBatch in one launch.
     q: [num_prefill_tokens, Hq, D]."""
-    return _impl(q).attention_prefill_batch(q, k_cache, v_cache, batch, scale)
+    return _impl(q).attention_prefill_batch(q, k_cache, v_cache, batch, scale, out)
 
 
-def attention_decode_batch(q, k_cache, v_cache, batch, scale):
+def attention_decode_batch(q, k_cache, v_cache, batch, scale, out=None):
     """One-token-per-sequence paged attention.  q: [B, Hq, D]."""
-    return _impl(q).attention_decode_batch(q, k_cache, v_cache, batch, scale)
+    return _impl(q).attention_decode_batch(q, k_cache, v_cache, batch, scale, out)
 
 
 def swiglu(gate_up):

@@ -56,8 +56,9 @@ def _attn_out(q):
     return torch.empty(q.shape, dtype=q.dtype, device=q.device)
 
 
-def attention_prefill_batch(q, k_cache, v_cache, batch, scale):
-    out = _attn_out(q)
+def attention_prefill_batch(q, k_cache, v_cache, batch, scale, out=None):
+    if out is None:
+        out = _attn_out(q)
     if _PREFILL_IMPL == "mfma":
         meta = batch.prefill_meta(tile_q=128)
         _C.prefill_attn_mfma(
@@ -73,8 +74,9 @@ def attention_prefill_batch(q, k_cache, v_cache, batch, scale):
     return out
 
 
-def attention_decode_batch(q, k_cache, v_cache, batch, scale):
-    out = _attn_out(q)
+def attention_decode_batch(q, k_cache, v_cache, batch, scale, out=None):
+    if out is None:
+        out = _attn_out(q)
     _C.decode_attn(
         out, q, k_cache, v_cache, batch.decode_tables_i32(), batch.decode_lens_i32(),
         scale,

@@ -84,7 +84,7 @@ def attention_prefill(q, k_cache, v_cache, block_table, seq_len, ctx_len, scale)
     return out.to(q.dtype)
 
 
-def attention_prefill_batch(q, k_cache, v_cache, batch, scale):
+def attention_prefill_batch(q, k_cache, v_cache, batch, scale, out=None):
     """Loop the per-sequence oracle over a FlatBatch's prefill chunks."""
     outs = []
     row = 0
@@ -96,14 +96,22 @@ def attention_prefill_batch(q, k_cache, v_cache, batch, scale):
             )
         )
         row += m.query_len
-    return torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
+    res = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
+    if out is not None:
+        out.copy_(res)
+        return out
+    return res
 
 
-def attention_decode_batch(q, k_cache, v_cache, batch, scale):
+def attention_decode_batch(q, k_cache, v_cache, batch, scale, out=None):
     """q: [B, Hq, D]; decode tables/lens come from the FlatBatch."""
-    return attention_decode_raw(
+    res = attention_decode_raw(
         q, k_cache, v_cache, batch.decode_block_tables, batch.decode_seq_lens, scale
     )
+    if out is not None:
+        out.copy_(res)
+        return out
+    return res
 
 
 def attention_decode_raw(q, k_cache, v_cache, block_tables, seq_lens, scale):
