@@ -63,10 +63,12 @@ def _tp_forward_worker(rank, world, port, q):
         want = full.forward(mk())
         got = shard.forward(mk())
         err = (want - got).abs().max().item()
-        # KV caches must also agree on this rank's kv-head slice
+        # KV caches must agree on this rank's kv-head slice (written slots
+        # only — the pool is uninitialized by design)
         kv_per = cfg.num_kv_heads // world
         kv_err = (
-            (full.k_caches[0][:, :, rank * kv_per : (rank + 1) * kv_per] - shard.k_caches[0])
+            (full.k_caches[0][0, :T, rank * kv_per : (rank + 1) * kv_per]
+             - shard.k_caches[0][0, :T])
             .abs().max().item()
         )
         q.put((rank, err, kv_err))
