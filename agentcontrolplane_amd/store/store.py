@@ -80,6 +80,11 @@ class ResourceStore:
         self._lock = threading.RLock()
         # kind -> namespace -> name -> obj
         self._data: Dict[str, Dict[str, Dict[str, Dict[str, Any]]]] = {}
+        # owner uid -> [(kind, ns, name)] — O(children) cascade deletes
+        self._owned_by: Dict[str, List[Tuple[str, str, str]]] = {}
+        # per-namespace FIFO of event names for TTL-style capping
+        self._event_fifo: Dict[str, List[str]] = {}
+        self.max_events_per_namespace = 20000
         self._rv = 0
         self._watches: List[_Watch] = []
         self._wal_path = wal_path
@@ -119,6 +124,16 @@ class ResourceStore:
                 else:
                     bucket[name] = obj
                 self._rv = max(self._rv, int(m.get("resourceVersion", 0)))
+        # rebuild the owner index from the replayed state
+        self._owned_by.clear()
+        for kind, nss in self._data.items():
+            for ns, objs in nss.items():
+                for name, obj in objs.items():
+                    for ref in obj.get("metadata", {}).get("ownerReferences", []) or []:
+                        if ref.get("uid"):
+                            self._owned_by.setdefault(ref["uid"], []).append(
+                                (kind, ns, name)
+                            )
 
     def _append_wal(self, op: str, obj: Dict[str, Any]) -> None:
         if self._wal_file is None:
@@ -188,6 +203,9 @@ class ResourceStore:
             m["resourceVersion"] = self._rv
             m["generation"] = 1
             bucket[name] = obj
+            for ref in m.get("ownerReferences", []) or []:
+                if ref.get("uid"):
+                    self._owned_by.setdefault(ref["uid"], []).append((kind, ns, name))
             self._append_wal("put", obj)
             out = copy.deepcopy(obj)
             self._notify(WatchEvent("ADDED", kind, out))
@@ -277,16 +295,11 @@ class ResourceStore:
                 return False
             self._append_wal("delete", obj)
             self._notify(WatchEvent("DELETED", kind, copy.deepcopy(obj)))
-            # cascade: delete objects owned by this uid (k8s GC role)
+            # cascade via the owner index (k8s GC role), O(children)
             uid = obj.get("metadata", {}).get("uid")
             if uid:
-                for k2, nss in list(self._data.items()):
-                    for ns2, objs in list(nss.items()):
-                        for n2, o2 in list(objs.items()):
-                            for ref in o2.get("metadata", {}).get("ownerReferences", []) or []:
-                                if ref.get("uid") == uid:
-                                    self.delete(k2, n2, ns2)
-                                    break
+                for (k2, ns2, n2) in self._owned_by.pop(uid, []):
+                    self.delete(k2, n2, ns2)
         return True
 
     # --------------------------------------------------------------- watches
@@ -351,12 +364,16 @@ class ResourceStore:
             "lastTimestamp": now_iso(),
         }
         with self._lock:
-            bucket = self._data.setdefault(EVENT, {}).setdefault(
-                m.get("namespace", "default"), {}
-            )
+            ns = m.get("namespace", "default")
+            bucket = self._data.setdefault(EVENT, {}).setdefault(ns, {})
             self._rv += 1
             ev["metadata"]["resourceVersion"] = self._rv
             bucket[ev["metadata"]["name"]] = ev
+            fifo = self._event_fifo.setdefault(ns, [])
+            fifo.append(ev["metadata"]["name"])
+            # TTL-style cap (k8s events expire after 1h; here: count-bound)
+            while len(fifo) > self.max_events_per_namespace:
+                bucket.pop(fifo.pop(0), None)
             self._append_wal("put", ev)
             self._notify(WatchEvent("ADDED", EVENT, copy.deepcopy(ev)))
 
