@@ -29,9 +29,20 @@ class PyBlockManager:
         self._free: List[int] = list(range(num_blocks - 1, -1, -1))
         self._tables: Dict[int, List[int]] = {}
         self._lens: Dict[int, int] = {}
+        # copy-on-nothing sharing: a block referenced by >1 table is
+        # immutable (full) and freed when its count drops to 0
+        self._refs: Dict[int, int] = {}
         #: bumps whenever any block table changes shape — callers cache
         #: derived tensors against it
         self.table_epoch = 0
+
+    def _release_block(self, b: int) -> None:
+        n = self._refs.get(b, 1) - 1
+        if n <= 0:
+            self._refs.pop(b, None)
+            self._free.append(b)
+        else:
+            self._refs[b] = n
 
     # ------------------------------------------------------------ lifecycle
 
@@ -45,7 +56,8 @@ class PyBlockManager:
         blocks = self._tables.pop(seq_id, None)
         self._lens.pop(seq_id, None)
         if blocks:
-            self._free.extend(reversed(blocks))
+            for b in reversed(blocks):
+                self._release_block(b)
             self.table_epoch += 1
 
     def has_seq(self, seq_id: int) -> bool:
@@ -98,9 +110,30 @@ class PyBlockManager:
         assert n_blocks <= len(old) and n_tokens <= n_blocks * self.block_size
         self._tables[new_seq_id] = old[:n_blocks]
         self._lens[new_seq_id] = n_tokens
-        if len(old) > n_blocks:
-            self._free.extend(reversed(old[n_blocks:]))
+        for b in reversed(old[n_blocks:]):
+            self._release_block(b)
         self.table_epoch += 1
+
+    def share_prefix(self, new_seq_id: int, donor_seq_id: int, n_blocks: int,
+                     n_tokens: int) -> None:
+        """Register a NEW sequence whose first n_blocks alias the donor's
+        (refcounted; the shared blocks are full and immutable — appends only
+        ever touch blocks past n_tokens).  The donor keeps its table."""
+        if new_seq_id in self._tables:
+            raise KeyError(f"seq {new_seq_id} already exists")
+        donor = self._tables.get(donor_seq_id)
+        if donor is None:
+            raise KeyError(f"seq {donor_seq_id} not found")
+        assert n_blocks <= len(donor) and n_tokens == n_blocks * self.block_size
+        shared = donor[:n_blocks]
+        for b in shared:
+            self._refs[b] = self._refs.get(b, 1) + 1
+        self._tables[new_seq_id] = list(shared)
+        self._lens[new_seq_id] = n_tokens
+        self.table_epoch += 1
+
+    def ref_count(self, block: int) -> int:
+        return self._refs.get(block, 1)
 
     # -------------------------------------------------------------- queries
 

@@ -336,15 +336,24 @@ class Scheduler:
         while self.bm.free_blocks < need_blocks and self._evict_one_retired():
             pass
 
+    #: newest candidates compared per admission (bounds the token compares
+    #: when many conversations share a first block, e.g. a common system
+    #: prompt across 1k tasks)
+    ADOPT_CANDIDATES = 8
+
     def _try_adopt(self, seq: Sequence) -> bool:
-        """Adopt a retired conversation's KV prefix; returns True if the
-        sequence was registered with the block manager."""
+        """Share a retired conversation's KV prefix (refcounted, full blocks
+        only); returns True if the sequence was registered with the block
+        manager.  The retiree stays cached so any number of requests with
+        the same prefix — the next turn of the same task, or sibling tasks
+        sharing a system prompt — reuse it."""
         bs = self.cfg.kv_block_size
         if not self.continuation_cache or len(seq.prompt_ids) <= bs:
             return False
         key = self._first_block_key(seq.prompt_ids)
+        bucket = self._retired_index.get(key, [])
         best_id, best_len = None, 0
-        for cand in self._retired_index.get(key, []):
+        for cand in reversed(bucket[-self.ADOPT_CANDIDATES :]):
             toks = self.retired[cand]
             limit = min(len(toks), len(seq.prompt_ids) - 1)
             n = bs  # first block already known equal
@@ -355,13 +364,7 @@ class Scheduler:
         n_blocks = (best_len // bs) if best_id is not None else 0
         if n_blocks < 1:
             return False
-        toks = self.retired.pop(best_id)
-        bucket = self._retired_index.get(key)
-        if bucket is not None:
-            bucket.remove(best_id)
-            if not bucket:
-                self._retired_index.pop(key, None)
-        self.bm.adopt_prefix(seq.seq_id, best_id, n_blocks, n_blocks * bs)
+        self.bm.share_prefix(seq.seq_id, best_id, n_blocks, n_blocks * bs)
         seq.num_processed = n_blocks * bs
         self.continuation_hits += 1
         self.continuation_tokens_saved += n_blocks * bs

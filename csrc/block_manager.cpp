@@ -34,7 +34,7 @@ class BlockManager {
     if (it == tables_.end()) return;
     if (!it->second.empty()) ++table_epoch_;
     for (auto rit = it->second.rbegin(); rit != it->second.rend(); ++rit)
-      free_.push_back(*rit);
+      release_block(*rit);
     tables_.erase(it);
     lens_.erase(seq_id);
   }
@@ -84,8 +84,32 @@ class BlockManager {
         std::vector<int64_t>(old.begin(), old.begin() + n_blocks);
     lens_[new_seq_id] = n_tokens;
     for (auto rit = old.rbegin(); rit != old.rend() - n_blocks; ++rit)
-      free_.push_back(*rit);
+      release_block(*rit);
     ++table_epoch_;
+  }
+
+  void share_prefix(int64_t new_seq_id, int64_t donor_seq_id, int64_t n_blocks,
+                    int64_t n_tokens) {
+    if (tables_.count(new_seq_id))
+      throw py::key_error("seq " + std::to_string(new_seq_id) + " already exists");
+    auto it = tables_.find(donor_seq_id);
+    if (it == tables_.end())
+      throw py::key_error("seq " + std::to_string(donor_seq_id) + " not found");
+    if (n_blocks > (int64_t)it->second.size() || n_tokens != n_blocks * block_size_)
+      throw std::runtime_error("share_prefix: full-block prefixes only");
+    std::vector<int64_t> shared(it->second.begin(), it->second.begin() + n_blocks);
+    for (int64_t b : shared) {
+      auto r = refs_.find(b);
+      refs_[b] = (r == refs_.end() ? 1 : r->second) + 1;
+    }
+    tables_[new_seq_id] = std::move(shared);
+    lens_[new_seq_id] = n_tokens;
+    ++table_epoch_;
+  }
+
+  int64_t ref_count(int64_t block) const {
+    auto r = refs_.find(block);
+    return r == refs_.end() ? 1 : r->second;
   }
 
   std::vector<int64_t> block_table(int64_t seq_id) const {
@@ -104,6 +128,16 @@ class BlockManager {
   }
 
  private:
+  void release_block(int64_t b) {
+    auto r = refs_.find(b);
+    if (r == refs_.end() || r->second <= 1) {
+      if (r != refs_.end()) refs_.erase(r);
+      free_.push_back(b);
+    } else {
+      --r->second;
+    }
+  }
+
   int64_t blocks_needed(int64_t seq_id, int64_t n_tokens) const {
     const int64_t cur = lens_.at(seq_id);
     const int64_t have = (int64_t)tables_.at(seq_id).size();
@@ -116,6 +150,7 @@ class BlockManager {
   std::vector<int64_t> free_;
   std::unordered_map<int64_t, std::vector<int64_t>> tables_;
   std::unordered_map<int64_t, int64_t> lens_;
+  std::unordered_map<int64_t, int64_t> refs_;
 };
 
 void register_block_manager(py::module_& m) {
@@ -128,6 +163,8 @@ void register_block_manager(py::module_& m) {
       .def("can_append", &BlockManager::can_append)
       .def("append_tokens", &BlockManager::append_tokens)
       .def("adopt_prefix", &BlockManager::adopt_prefix)
+      .def("share_prefix", &BlockManager::share_prefix)
+      .def("ref_count", &BlockManager::ref_count)
       .def("block_table", &BlockManager::block_table)
       .def("seq_len", &BlockManager::seq_len)
       .def("occupancy", &BlockManager::occupancy)

@@ -86,3 +86,54 @@ def test_block_manager_adopt_prefix_bookkeeping():
     # appends continue from the adopted prefix
     slots = bm.append_tokens(2, 1)
     assert slots[0] // 4 not in t1[:2] or slots[0] // 4 == t1[2]
+
+
+def test_shared_prefix_multiple_adopters():
+    """Sibling requests with the same prefix share the retiree's blocks
+    (refcounted) — all outputs identical to cold runs."""
+    eng = make_engine()
+    try:
+        base = list(range(50, 114))  # 64 tokens = 4 full blocks
+        r0 = eng.generate(base + [1, 2, 3], SamplingParams(max_tokens=4, temperature=0))
+        outs = []
+        for suffix in ([4, 5, 6], [7, 8, 9], [10, 11, 12]):
+            outs.append(
+                eng.generate(base + suffix, SamplingParams(max_tokens=4, temperature=0))
+            )
+        m = eng.metrics()
+        assert m["continuation_hits"] >= 3
+    finally:
+        eng.stop()
+    # each against a cold engine
+    for suffix, r in zip(([4, 5, 6], [7, 8, 9], [10, 11, 12]), outs):
+        eng2 = make_engine()
+        try:
+            cold = eng2.generate(base + suffix, SamplingParams(max_tokens=4, temperature=0))
+            assert r.output_ids == cold.output_ids, f"suffix {suffix}"
+        finally:
+            eng2.stop()
+
+
+def test_share_prefix_refcounts():
+    from agentcontrolplane_amd.engine.kv import PyBlockManager
+
+    bm = PyBlockManager(num_blocks=16, block_size=4)
+    bm.add_seq(1)
+    bm.append_tokens(1, 8)  # 2 blocks
+    t1 = bm.block_table(1)
+    bm.share_prefix(2, 1, 2, 8)
+    bm.share_prefix(3, 1, 2, 8)
+    assert bm.block_table(2) == t1 and bm.block_table(3) == t1
+    assert bm.free_blocks == 14
+    bm.free_seq(1)
+    assert bm.free_blocks == 14  # still referenced
+    bm.free_seq(2)
+    assert bm.free_blocks == 14
+    bm.free_seq(3)
+    assert bm.free_blocks == 16
+    # appends after sharing go to fresh blocks
+    bm.add_seq(9)
+    bm.append_tokens(9, 8)
+    bm.share_prefix(10, 9, 2, 8)
+    slots = bm.append_tokens(10, 1)
+    assert slots[0] // 4 not in bm.block_table(9)
