@@ -85,7 +85,7 @@ class ControlPlane:
         if self._rest_app is None:
             from .server.rest import build_app
 
-            self._rest_app = build_app(self.store, self.manager)
+            self._rest_app = build_app(self.store, self.manager, engine=self.engine)
         return self._rest_app
 
     # ------------------------------------------------------------- lifecycle

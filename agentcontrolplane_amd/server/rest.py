@@ -327,6 +327,81 @@ def build_app(store, manager=None, engine=None) -> FastAPI:
             )
         return {"id": call_id, "resolved": True}
 
+    # ------------------------------------------------- OpenAI-compatible API
+
+    @app.get("/v1/models")
+    def list_models():
+        models = []
+        if engine is not None:
+            models.append(
+                {
+                    "id": engine.cfg.model,
+                    "object": "model",
+                    "owned_by": "agentcontrolplane_amd",
+                }
+            )
+        return {"object": "list", "data": models}
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(request: Request):
+        """OpenAI-wire chat completion against the in-process engine, so
+        existing OpenAI clients (and the reference's mock-server-based e2e
+        suites) can point straight at this server."""
+        if engine is None:
+            return JSONResponse({"error": "no engine attached"}, status_code=503)
+        body = json.loads(await request.body())
+        from ..engine.request import SamplingParams
+
+        messages = body.get("messages", [])
+        # OpenAI tool_calls → internal toolCalls rendering
+        for m in messages:
+            if m.get("tool_calls"):
+                m["toolCalls"] = m.pop("tool_calls")
+        tools = body.get("tools", []) or []
+        tool_choice = body.get("tool_choice", "auto")
+        sampling = SamplingParams(
+            max_tokens=int(body.get("max_tokens") or 256),
+            temperature=float(body.get("temperature", 0.7)),
+            top_p=float(body.get("top_p", 1.0)),
+            tool_choice=tool_choice if isinstance(tool_choice, str) else "required",
+        )
+        try:
+            import anyio
+
+            result = await anyio.to_thread.run_sync(
+                lambda: engine.chat(messages, tools, sampling)
+            )
+        except ValueError as e:
+            return JSONResponse(
+                {"error": {"message": str(e), "type": "invalid_request_error"}},
+                status_code=400,
+            )
+        except Exception as e:  # noqa: BLE001
+            return JSONResponse(
+                {"error": {"message": str(e), "type": "server_error"}}, status_code=500
+            )
+        msg: Dict[str, Any] = {"role": "assistant", "content": result.text or None}
+        finish = "stop" if result.finish_reason in ("stop", "length") else result.finish_reason
+        if result.tool_calls:
+            msg["tool_calls"] = result.tool_calls
+            msg["content"] = None
+            finish = "tool_calls"
+        import time as _time
+        import uuid as _uuid
+
+        return {
+            "id": f"chatcmpl-{_uuid.uuid4().hex[:24]}",
+            "object": "chat.completion",
+            "created": int(_time.time()),
+            "model": body.get("model", engine.cfg.model),
+            "choices": [{"index": 0, "message": msg, "finish_reason": finish}],
+            "usage": {
+                "prompt_tokens": result.prompt_tokens,
+                "completion_tokens": result.completion_tokens,
+                "total_tokens": result.prompt_tokens + result.completion_tokens,
+            },
+        }
+
     # ---------------------------------------------------------------- metrics
 
     @app.get("/metrics")

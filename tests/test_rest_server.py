@@ -169,3 +169,55 @@ def test_metrics_endpoint(client, cp):
     r = client.get("/metrics")
     assert r.status_code == 200
     assert "acp_up 1" in r.text
+
+
+def test_openai_chat_completions_endpoint():
+    """OpenAI-wire completions against the in-process engine."""
+    from agentcontrolplane_amd.engine.config import EngineConfig
+    from agentcontrolplane_amd.engine.engine import InferenceEngine
+
+    eng = InferenceEngine(
+        EngineConfig(model="tiny", device="cpu", num_kv_blocks=1024, kv_block_size=16,
+                     max_prefill_tokens=256, request_timeout_s=120)
+    )
+    plane = ControlPlane(engine=eng, auto_approve="approve", llm_probe=False)
+    plane.start()
+    try:
+        client = TestClient(plane.rest_app)
+        r = client.get("/v1/models")
+        assert r.json()["data"][0]["id"] == "tiny"
+        r = client.post(
+            "/v1/chat/completions",
+            json={
+                "model": "tiny",
+                "messages": [{"role": "user", "content": "hi"}],
+                "max_tokens": 8,
+                "temperature": 0.9,
+            },
+        )
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert body["choices"][0]["finish_reason"] == "stop"
+        assert body["usage"]["completion_tokens"] <= 8
+        # tool-calling turn
+        r = client.post(
+            "/v1/chat/completions",
+            json={
+                "model": "tiny",
+                "messages": [{"role": "user", "content": "use the tool"}],
+                "tools": [{"type": "function",
+                           "function": {"name": "calc__add", "parameters": {}}}],
+                "tool_choice": "required",
+                "max_tokens": 48,
+            },
+        )
+        body = r.json()
+        choice = body["choices"][0]
+        assert choice["finish_reason"] == "tool_calls"
+        assert choice["message"]["tool_calls"][0]["function"]["name"] == "calc__add"
+        import json as _json
+
+        _json.loads(choice["message"]["tool_calls"][0]["function"]["arguments"])
+    finally:
+        plane.stop()
+        eng.stop()
