@@ -97,12 +97,13 @@ def test_engine_error_mid_task_retries():
     still completes (reference handleLLMError semantics)."""
     from agentcontrolplane_amd.llmclient.base import LLMRequestError
 
-    calls = {"n": 0}
+    failed = {"done": False}
 
     class Flaky(MockLLMClient):
         def send_request(self, messages, tools):
-            calls["n"] += 1
-            if calls["n"] == 2:  # fail the first task turn after the probe
+            is_probe = len(messages) == 1 and messages[0].content == "ping"
+            if not is_probe and not failed["done"]:
+                failed["done"] = True  # fail exactly the first real task turn
                 raise LLMRequestError(503, "transient engine failure")
             return super().send_request(messages, tools)
 
