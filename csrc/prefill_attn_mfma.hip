@@ -58,9 +58,10 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
 
   // LDS: K row-major [32][128] with XOR swizzle on 16-B units to break the
   // ds_read_b128 16-way conflict (guide G4: byte ^= (row&7)<<4); V
-  // transposed [128][32] (row = d, 64 B) read conflict-light.
-  __shared__ bf16_t Kt[KVBLK][D_HEAD];
-  __shared__ bf16_t Vt[D_HEAD][KVBLK];
+  // transposed [128][32] (row = d, 64 B) read conflict-light.  Double
+  // buffered: tile t+1 stages while tile t computes (one barrier/tile).
+  __shared__ bf16_t Kt[2][KVBLK][D_HEAD];
+  __shared__ bf16_t Vt[2][D_HEAD][KVBLK];
   __shared__ float bcast[4][QBLK];
   __shared__ int bt[512];
   const int nblk = (S + kv_block - 1) / kv_block;
@@ -88,39 +89,43 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
   f32x16_v o0 = {}, o1 = {}, o2 = {}, o3 = {};
 
   const int k_end = min(S, ctx + q0 + rows_here);
-  for (int kv = 0; kv < k_end; kv += KVBLK) {
-    const int kn = min(KVBLK, k_end - kv);
-    __syncthreads();
-    // stage K (swizzled 16-B units) and V^T
-    {
-      // 32x128 bf16 = 512 16-B pieces; 256 threads x 2
-      for (int i = threadIdx.x; i < (KVBLK * D_HEAD) / 8; i += blockDim.x) {
-        const int kk = i >> 4;           // key row (128 d / 8 = 16 pieces per row)
-        const int dd = (i & 15) << 3;    // d offset
-        const int j = kv + kk;
-        BF16x8 v8;
-        if (kk < kn) {
-          const int64_t slot = (int64_t)bt[j / kv_block] * kv_block + j % kv_block;
-          v8 = *(const BF16x8*)(k_cache + (slot * Hkv + kvh) * D_HEAD + dd);
-        } else {
-          v8.u128 = ulonglong2{0, 0};
-        }
-        // swizzled K store: 16-B unit index dd/8 XORed with row&7
-        const int sw = (dd >> 3) ^ (kk & 7);
-        *(BF16x8*)(&Kt[kk][sw << 3]) = v8;
-        // V^T scatter (8 x 2-B stores)
-        BF16x8 vv;
-        if (kk < kn) {
-          const int64_t slot = (int64_t)bt[j / kv_block] * kv_block + j % kv_block;
-          vv = *(const BF16x8*)(v_cache + (slot * Hkv + kvh) * D_HEAD + dd);
-        } else {
-          vv.u128 = ulonglong2{0, 0};
-        }
-#pragma unroll
-        for (int e = 0; e < 8; ++e) Vt[dd + e][kk] = vv.h[e];
+  const int n_tiles = (k_end + KVBLK - 1) / KVBLK;
+
+  auto stage_tile = [&](int tile_idx, int buf) {
+    const int kv0 = tile_idx * KVBLK;
+    const int kn0 = min(KVBLK, k_end - kv0);
+    // 32x128 bf16 = 512 16-B pieces; 256 threads x 2
+    for (int i = threadIdx.x; i < (KVBLK * D_HEAD) / 8; i += blockDim.x) {
+      const int kk = i >> 4;           // key row (128 d / 8 = 16 pieces per row)
+      const int dd = (i & 15) << 3;    // d offset
+      const int j = kv0 + kk;
+      BF16x8 v8, vv;
+      if (kk < kn0) {
+        const int64_t slot = (int64_t)bt[j / kv_block] * kv_block + j % kv_block;
+        v8 = *(const BF16x8*)(k_cache + (slot * Hkv + kvh) * D_HEAD + dd);
+        vv = *(const BF16x8*)(v_cache + (slot * Hkv + kvh) * D_HEAD + dd);
+      } else {
+        v8.u128 = ulonglong2{0, 0};
+        vv.u128 = ulonglong2{0, 0};
       }
+      // swizzled K store: 16-B unit index dd/8 XORed with row&7
+      const int sw = (dd >> 3) ^ (kk & 7);
+      *(BF16x8*)(&Kt[buf][kk][sw << 3]) = v8;
+      // V^T scatter (8 x 2-B stores)
+#pragma unroll
+      for (int e = 0; e < 8; ++e) Vt[buf][dd + e][kk] = vv.h[e];
     }
-    __syncthreads();
+  };
+
+  __syncthreads();
+  if (n_tiles > 0) stage_tile(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv = t * KVBLK;
+    const int kn = min(KVBLK, k_end - kv);
+    // stage the NEXT tile into the other buffer while computing this one
+    if (t + 1 < n_tiles) stage_tile(t + 1, cur ^ 1);
 
     // QK^T: S_tile[k][q] = sum_d K[k][d] * Q[q][d]
     f32x16_v s = {};
@@ -128,7 +133,7 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
     for (int kk = 0; kk < 8; ++kk) {
       // A-frag: lane holds K[row = lane&31][d = kk*16 + khalf*8 + e]
       const int unit = ((kk * 16 + khalf * 8) >> 3) ^ ((lane & 31) & 7);
-      bf16x8_v a = *(const bf16x8_v*)(&Kt[lane & 31][unit << 3]);
+      bf16x8_v a = *(const bf16x8_v*)(&Kt[cur][lane & 31][unit << 3]);
       s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qfrag[kk], s, 0, 0, 0);
     }
 
@@ -198,13 +203,15 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
     //         = Vt[dtile*32 + (lane&31)][k…] — 16 B contiguous
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
-      const bf16_t* vrow = &Vt[dt * 32 + (lane & 31)][khalf * 8];
+      const bf16_t* vrow = &Vt[cur][dt * 32 + (lane & 31)][khalf * 8];
       bf16x8_v b_lo = *(const bf16x8_v*)(vrow);
       bf16x8_v b_hi = *(const bf16x8_v*)(vrow + 16);
       f32x16_v* od = dt == 0 ? &o0 : dt == 1 ? &o1 : dt == 2 ? &o2 : &o3;
       *od = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa_lo, b_lo, *od, 0, 0, 0);
       *od = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa_hi, b_hi, *od, 0, 0, 0);
     }
+    __syncthreads();  // staging of t+1 done AND everyone done reading cur
+    cur ^= 1;
   }
 
   // epilogue: normalize by 1/l_run per q (broadcast through LDS), store
