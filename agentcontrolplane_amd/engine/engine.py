@@ -67,7 +67,10 @@ class InferenceEngine:
         )
         self.scheduler = Scheduler(config, self.bm, self.device)
         self.graph_runner = None
-        if self.device.type == "cuda" and not config.enforce_eager:
+        # MoE forward has data-dependent expert dispatch (token→expert
+        # routing changes per step) — capture would freeze one routing, so
+        # graphs are dense-model only
+        if self.device.type == "cuda" and not config.enforce_eager and not self.mcfg.is_moe:
             from .graphs import DecodeGraphRunner
 
             max_blocks_per_seq = (
