@@ -133,6 +133,12 @@ def main() -> None:
     p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree (else DP sharding)")
     p.add_argument("--device", default=None)
     p.add_argument("--kv-blocks", type=int, default=None)
+    p.add_argument("--prefill-budget", type=int, default=8192)
+    p.add_argument(
+        "--approval-gate", action="store_true",
+        help="route the MCP tool through a human-approval ContactChannel "
+        "(BASELINE.json config 5's gate; approvals auto-resolve)",
+    )
     p.add_argument(
         "--no-wal", action="store_true",
         help="disable the durable WAL (default on: the reference checkpoints "
@@ -166,7 +172,7 @@ def main() -> None:
         model=model,
         device=device,
         num_kv_blocks=args.kv_blocks if args.kv_blocks else (1024 if device == "cpu" else None),
-        max_prefill_tokens=8192,
+        max_prefill_tokens=args.prefill_budget,
         tensor_parallel=args.tp,
         seed=1234 + rank,
         request_timeout_s=1200,
@@ -214,7 +220,26 @@ def main() -> None:
         llm["status"].update({"ready": True, "status": "Ready"})
         cp.store.update_status(llm)
         cp.mcp.register_inproc("tools", {"noop": lambda **_: "ok"})
-        cp.store.create(make_resource(MCP_SERVER, "tools", spec={"transport": "inproc"}))
+        mcp_spec = {"transport": "inproc"}
+        if args.approval_gate:
+            cp.store.create(
+                make_resource(
+                    "Secret", "hl-key", spec={"data": {"k": "hl-bench"}}, api_version="v1"
+                )
+            )
+            cp.store.create(
+                make_resource(
+                    "ContactChannel",
+                    "bench-approvals",
+                    spec={
+                        "type": "slack",
+                        "apiKeyFrom": {"secretKeyRef": {"name": "hl-key", "key": "k"}},
+                        "slack": {"channelOrUserID": "CBENCH"},
+                    },
+                )
+            )
+            mcp_spec["approvalContactChannel"] = {"name": "bench-approvals"}
+        cp.store.create(make_resource(MCP_SERVER, "tools", spec=mcp_spec))
         cp.store.create(
             make_resource(
                 AGENT,
