@@ -300,6 +300,148 @@ class JsonValueMachine:
             self.state = "arr_next"
 
 
+class SchemaArgsMachine:
+    """Arguments object constrained to a simple JSON schema: the required
+    properties are emitted in order with type-correct values, so the tool
+    call is not just parseable but *executable* (e.g. delegate_to_agent's
+    required ``message`` string, calculator's numeric ``a``/``b``).
+
+    Supports property types string / number / integer / boolean; anything
+    richer falls back to the free JsonValueMachine at the value position.
+    """
+
+    def __init__(self, schema: dict, max_len: int = 256):
+        props = (schema or {}).get("properties", {}) or {}
+        required = (schema or {}).get("required", []) or []
+        keys = [k for k in required if k in props] or list(props.keys())[:1]
+        self.fields = [(k, (props.get(k) or {}).get("type", "string")) for k in keys]
+        self.max_len = max_len
+        self.count = 0
+        self.done = False
+        # build the program: list of ("lit", bytes) | ("str",) | ("num",) |
+        # ("bool",) | ("free",)
+        prog: List[tuple] = [("lit", b"{")]
+        for i, (k, t) in enumerate(self.fields):
+            sep = b", " if i else b""
+            prog.append(("lit", sep + b'"' + k.encode() + b'": '))
+            if t == "string":
+                prog.append(("str",))
+            elif t in ("number", "integer"):
+                prog.append(("num",))
+            elif t == "boolean":
+                prog.append(("bool",))
+            else:
+                prog.append(("free",))
+        prog.append(("lit", b"}"))
+        self.prog = prog
+        self.pi = 0       # program counter
+        self.off = 0      # offset within a literal / value state
+        self.sub: Optional[JsonValueMachine] = None
+        self._bool_rest = b""
+
+    def _cur(self):
+        return self.prog[self.pi] if self.pi < len(self.prog) else None
+
+    def allowed(self) -> Set[int]:
+        cur = self._cur()
+        if cur is None:
+            return set()
+        kind = cur[0]
+        if kind == "lit":
+            return {cur[1][self.off]}
+        budget_left = self.count < self.max_len
+        if kind == "str":
+            if self.off == 0:
+                return {0x22}
+            opts = {0x22}
+            if budget_left:
+                opts |= set(STRING_SAFE)
+            return opts
+        if kind == "num":
+            if self.off == 0:
+                return _bs(b"-123456789") | {ord("0")}
+            if self.off == -2:  # after '-': a first digit is mandatory
+                return _bs(DIGITS)
+            if self.off == -1:  # after a leading zero: number is complete
+                return self._next_lit_first()
+            return _bs(DIGITS) | self._next_lit_first() if budget_left else self._next_lit_first()
+        if kind == "bool":
+            if self.off == 0:
+                return _bs(b"tf")
+            return {self._bool_rest[0]}
+        if kind == "free":
+            if self.sub is None:
+                # object-rooted: a root-level scalar has no terminator of its
+                # own, so untyped/object-typed fields emit a JSON object
+                self.sub = JsonValueMachine(
+                    max_len=max(16, self.max_len - self.count), root_object=True
+                )
+            opts = set(self.sub.allowed())
+            if self.sub.done:
+                opts |= self._next_lit_first()
+            return opts
+        raise AssertionError(kind)
+
+    def _next_lit_first(self) -> Set[int]:
+        nxt = self.prog[self.pi + 1]
+        return {nxt[1][0]}
+
+    def advance(self, b: int) -> None:
+        self.count += 1
+        cur = self._cur()
+        kind = cur[0]
+        if kind == "lit":
+            self.off += 1
+            if self.off == len(cur[1]):
+                self.pi += 1
+                self.off = 0
+                if self.pi == len(self.prog):
+                    self.done = True
+            return
+        if kind == "str":
+            if self.off > 0 and b == 0x22:
+                self.pi += 1
+                self.off = 0
+                return
+            self.off += 1
+            return
+        if kind == "num":
+            if (self.off > 0 or self.off == -1) and b not in DIGITS:
+                # terminator: belongs to the next literal
+                self.pi += 1
+                self.off = 0
+                self.count -= 1
+                return self.advance(b)
+            if self.off == 0:
+                self.off = -2 if b == ord("-") else (-1 if b == ord("0") else 1)
+                return
+            if self.off == -2:
+                self.off = -1 if b == ord("0") else 1
+                return
+            self.off += 1
+            return
+        if kind == "bool":
+            if self.off == 0:
+                self._bool_rest = b"rue" if b == ord("t") else b"alse"
+                self.off = 1
+                return
+            self._bool_rest = self._bool_rest[1:]
+            if not self._bool_rest:
+                self.pi += 1
+                self.off = 0
+            return
+        if kind == "free":
+            if self.sub.done and b in self._next_lit_first():
+                self.pi += 1
+                self.off = 0
+                self.sub = None
+                self.count -= 1
+                return self.advance(b)
+            self.sub.advance(b)
+            return
+        raise AssertionError(kind)
+
+
 class ToolCallGrammar:
     """Constrains one tool call:
     ``{"name": "<tool>", "arguments": {…}}`` then EOT.
@@ -312,16 +454,36 @@ class ToolCallGrammar:
     PRE = b'{"name": "'
     MID = b'", "arguments": '
 
-    def __init__(self, tool_names: List[str], max_args_len: int = 2048):
+    def __init__(self, tool_names: Optional[List[str]] = None, max_args_len: int = 2048,
+                 tools: Optional[List[dict]] = None):
+        self.schemas: Dict[str, dict] = {}
+        if tools:
+            names = []
+            for t in tools:
+                fn = t.get("function", {}) or {}
+                if fn.get("name"):
+                    names.append(fn["name"])
+                    self.schemas[fn["name"]] = fn.get("parameters") or {}
+            tool_names = names
         if not tool_names:
             raise ValueError("no tools to constrain to")
         self.names = sorted(set(tool_names))
         self.buf = bytearray()
         self.phase = "pre"       # pre → name → mid → args → done
         self.pos = 0
+        self.max_args_len = max_args_len
         self.name_prefix = b""
         self.args = JsonValueMachine(max_len=max_args_len, root_object=True)
         self.finished = False
+
+    def _select_args_machine(self, name: str) -> None:
+        """Once the tool is known, constrain the arguments to its parameter
+        schema so the produced call is executable, not merely parseable."""
+        schema = self.schemas.get(name) or {}
+        if schema.get("properties"):
+            self.args = SchemaArgsMachine(schema, max_len=self.max_args_len)
+        else:
+            self.args = JsonValueMachine(max_len=self.max_args_len, root_object=True)
 
     def allowed_tokens(self) -> Set[int]:
         if self.phase == "pre":
@@ -360,6 +522,7 @@ class ToolCallGrammar:
             return
         if self.phase == "name":
             if b == ord('"') and self.name_prefix.decode() in self.names:
+                self._select_args_machine(self.name_prefix.decode())
                 self.phase = "mid"
                 self.pos = 1  # the '"' consumed is MID[0]... MID starts with '"'
                 return

@@ -50,7 +50,7 @@ class Sequence:
             # the grammar's closing mode needs ~20 tokens of slack for the
             # name/scaffolding plus the shortest legal unwind
             max_args = max(16, request.sampling.max_tokens - 24 - max(len(n) for n in names))
-            self.grammar = ToolCallGrammar(names, max_args_len=max_args)
+            self.grammar = ToolCallGrammar(tools=request.tools, max_args_len=max_args)
 
     @property
     def total_len(self) -> int:

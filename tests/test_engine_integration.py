@@ -84,3 +84,54 @@ def test_agent_loop_on_local_engine(cp):
     json.loads(tc["function"]["arguments"])
     # the tool actually executed (result is a float string or an error-free str)
     assert cw[3]["content"] != ""
+
+
+def test_sub_agent_delegation_via_engine(cp):
+    """Config 4 at test scale: the ENGINE's constrained decoding picks among
+    delegate_to_agent__ tools; delegation spawns a child Task that runs its
+    own loop on the same engine."""
+    store = cp.store
+    store.create(
+        make_resource(
+            LLM,
+            "local-llm",
+            spec={
+                "provider": "local",
+                "parameters": {"model": "tiny", "maxTokens": 48, "temperature": "0.9"},
+            },
+        )
+    )
+    store.create(
+        make_resource(AGENT, "worker", spec={"llmRef": {"name": "local-llm"}, "system": "worker"})
+    )
+    store.create(
+        make_resource(
+            AGENT,
+            "lead",
+            spec={
+                "llmRef": {"name": "local-llm"},
+                "system": "lead",
+                "subAgents": [{"name": "worker"}],
+            },
+        )
+    )
+    store.create(
+        make_resource(TASK, "d1", spec={"agentRef": {"name": "lead"}, "userMessage": "do it"})
+    )
+    task = wait_for(
+        lambda: (store.get(TASK, "d1") or {}).get("status", {}).get("phase")
+        in (TaskPhase.FINAL_ANSWER, TaskPhase.FAILED)
+        and store.get(TASK, "d1"),
+        timeout=240,
+    )
+    assert task["status"]["phase"] == TaskPhase.FINAL_ANSWER, task["status"]
+    cw = task["status"]["contextWindow"]
+    # the only offered tool is the delegate tool → the constrained turn
+    # must have called it, and the child's answer is the tool result
+    tc = cw[2]["toolCalls"][0]
+    assert tc["function"]["name"] == "delegate_to_agent__worker"
+    children = [
+        t for t in store.list(TASK)
+        if (t["metadata"].get("labels") or {}).get("acp.humanlayer.dev/parent-toolcall")
+    ]
+    assert children and children[0]["status"]["phase"] == TaskPhase.FINAL_ANSWER

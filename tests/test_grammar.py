@@ -86,3 +86,47 @@ def test_json_depth_limit():
     for b in b'k": ':
         m.advance(b)
     assert ord("{") not in m.allowed() and ord("[") not in m.allowed()
+
+
+@pytest.mark.parametrize("seed", range(10))
+def test_schema_constrained_arguments(seed):
+    """Schema-aware args: required keys, correct types — executable calls."""
+    rng = random.Random(seed)
+    tools = [
+        {"type": "function", "function": {
+            "name": "delegate_to_agent__worker",
+            "parameters": {"type": "object",
+                           "properties": {"message": {"type": "string"}},
+                           "required": ["message"]}}},
+        {"type": "function", "function": {
+            "name": "calc__add",
+            "parameters": {"type": "object",
+                           "properties": {"a": {"type": "number"}, "b": {"type": "number"}},
+                           "required": ["a", "b"]}}},
+        {"type": "function", "function": {
+            "name": "flags__set",
+            "parameters": {"type": "object",
+                           "properties": {"on": {"type": "boolean"}},
+                           "required": ["on"]}}},
+    ]
+    g = ToolCallGrammar(tools=tools, max_args_len=80)
+    text = drive(g, rng)
+    obj = json.loads(text)
+    name, args = obj["name"], obj["arguments"]
+    if name == "delegate_to_agent__worker":
+        assert isinstance(args["message"], str)
+        assert set(args) == {"message"}
+    elif name == "calc__add":
+        assert isinstance(args["a"], (int, float)) and isinstance(args["b"], (int, float))
+        assert set(args) == {"a", "b"}
+    else:
+        assert isinstance(args["on"], bool)
+
+
+def test_schema_free_form_fallback():
+    """A tool without properties keeps the free-form object grammar."""
+    tools = [{"type": "function", "function": {"name": "t__x", "parameters": {}}}]
+    rng = random.Random(0)
+    g = ToolCallGrammar(tools=tools, max_args_len=60)
+    obj = json.loads(drive(g, rng))
+    assert isinstance(obj["arguments"], dict)
