@@ -135,6 +135,12 @@ def main() -> None:
     p.add_argument("--kv-blocks", type=int, default=None)
     p.add_argument("--prefill-budget", type=int, default=8192)
     p.add_argument(
+        "--sub-agents", type=int, default=0,
+        help="N child agents under the bench agent (BASELINE.json config 4: "
+        "the engine's constrained decoding may delegate; children run their "
+        "own full loops)",
+    )
+    p.add_argument(
         "--approval-gate", action="store_true",
         help="route the MCP tool through a human-approval ContactChannel "
         "(BASELINE.json config 5's gate; approvals auto-resolve)",
@@ -240,17 +246,28 @@ def main() -> None:
             )
             mcp_spec["approvalContactChannel"] = {"name": "bench-approvals"}
         cp.store.create(make_resource(MCP_SERVER, "tools", spec=mcp_spec))
-        cp.store.create(
-            make_resource(
-                AGENT,
-                "bench-agent",
-                spec={
-                    "llmRef": {"name": "bench-llm"},
-                    "system": "You are a benchmark agent. Use tools when offered.",
-                    "mcpServers": [{"name": "tools"}],
-                },
-            )
-        )
+        agent_spec = {
+            "llmRef": {"name": "bench-llm"},
+            "system": "You are a benchmark agent. Use tools when offered.",
+            "mcpServers": [{"name": "tools"}],
+        }
+        if args.sub_agents:
+            subs = []
+            for i in range(args.sub_agents):
+                cp.store.create(
+                    make_resource(
+                        AGENT,
+                        f"bench-child-{i}",
+                        spec={
+                            "llmRef": {"name": "bench-llm"},
+                            "system": f"You are specialist {i}.",
+                            "mcpServers": [{"name": "tools"}],
+                        },
+                    )
+                )
+                subs.append({"name": f"bench-child-{i}"})
+            agent_spec["subAgents"] = subs
+        cp.store.create(make_resource(AGENT, "bench-agent", spec=agent_spec))
         # wait for the agent to validate
         t0 = time.monotonic()
         while time.monotonic() - t0 < 60:
