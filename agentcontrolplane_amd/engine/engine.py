@@ -67,19 +67,33 @@ class InferenceEngine:
         )
         self.scheduler = Scheduler(config, self.bm, self.device)
         self.graph_runner = None
-        # MoE forward has data-dependent expert dispatch (token→expert
-        # routing changes per step) — capture would freeze one routing, so
-        # graphs are dense-model only
-        if self.device.type == "cuda" and not config.enforce_eager and not self.mcfg.is_moe:
+        # MoE decode batches under dense_moe_threshold take the all-experts
+        # dense path (static control flow) and are capture-safe; larger MoE
+        # batches (sparse routing loop) and EP dispatch stay eager
+        moe_graph_cap = None
+        if self.mcfg.is_moe:
+            if self.tp_world > 1 or getattr(self.model, "moe_dispatch", None) is not None:
+                moe_graph_cap = 0
+            else:
+                moe_graph_cap = getattr(self.model, "dense_moe_threshold", 0)
+        graphs_on = (
+            self.device.type == "cuda"
+            and not config.enforce_eager
+            and (moe_graph_cap is None or moe_graph_cap > 0)
+        )
+        if graphs_on:
             from .graphs import DecodeGraphRunner
 
             max_blocks_per_seq = (
                 min(config.max_model_len, self.mcfg.max_position)
                 + config.kv_block_size - 1
             ) // config.kv_block_size
+            max_graph_batch = min(config.max_batch_size, 512)
+            if moe_graph_cap is not None:
+                max_graph_batch = min(max_graph_batch, moe_graph_cap)
             self.graph_runner = DecodeGraphRunner(
                 self.model,
-                max_batch=min(config.max_batch_size, 512),
+                max_batch=max_graph_batch,
                 max_blocks_per_seq=max_blocks_per_seq,
                 scratch_block=num_blocks,
                 kv_block_size=config.kv_block_size,

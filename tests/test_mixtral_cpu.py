@@ -92,3 +92,19 @@ def test_moe_engine_end_to_end():
         assert res.tool_calls[0]["function"]["name"] == "t__x"
     finally:
         eng.stop()
+
+
+def test_dense_moe_path_matches_sparse():
+    """The all-experts dense (graph-capturable) decode path must equal the
+    sparse routing loop."""
+    torch.manual_seed(5)
+    cfg = PRESETS["tiny-moe"]
+    ecfg = EngineConfig(model="tiny-moe", device="cpu", num_kv_blocks=64)
+    model = MixtralForCausalLM(cfg, ecfg, "cpu")
+    model.random_init(0)
+    x = torch.randn(9, cfg.hidden_size)
+    model.dense_moe_threshold = 0
+    sparse = model._moe_mlp(0, x.clone())
+    model.dense_moe_threshold = 160
+    dense = model._moe_mlp(0, x.clone())
+    assert (sparse - dense).abs().max().item() < 1e-4
