@@ -299,7 +299,8 @@ def main() -> None:
             dist.barrier()
         elapsed = time.monotonic() - t_start
         if dp_sync:
-            t = torch.tensor([elapsed])
+            # nccl collectives need device tensors
+            t = torch.tensor([elapsed], device="cuda" if device == "cuda" else "cpu")
             dist.all_reduce(t, op=dist.ReduceOp.MAX)
             elapsed = float(t[0])
 
