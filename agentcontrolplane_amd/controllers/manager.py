@@ -24,7 +24,7 @@ import logging
 import threading
 import time
 import traceback
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Dict, List, Optional, Tuple
 
 from ..store import ResourceStore, WatchEvent
 

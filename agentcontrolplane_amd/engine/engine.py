@@ -13,7 +13,6 @@ grammar's legal next bytes each step.
 """
 from __future__ import annotations
 
-import json
 import threading
 import time
 from typing import Any, Dict, List, Optional
@@ -23,10 +22,9 @@ import torch
 from .. import ops
 from ..models import create_model
 from .config import EngineConfig
-from .grammar import ToolCallGrammar
 from .kv import make_block_manager
 from .request import ChatResult, InferenceRequest, SamplingParams
-from .scheduler import DECODE, Scheduler, Sequence
+from .scheduler import Scheduler, Sequence
 from .tokenizer import EOT, N_SPECIAL, TOOL_CALL_START, ByteTokenizer
 
 

@@ -23,7 +23,6 @@ import torch
 from .batch import FlatBatch, SeqMeta
 from .config import EngineConfig
 from .grammar import ToolCallGrammar
-from .kv import OutOfBlocksError
 from .request import InferenceRequest
 
 WAITING, PREFILL, DECODE, FINISHED = "waiting", "prefill", "decode", "finished"

@@ -13,7 +13,6 @@ of tasks, not the number of HTTP connections.
 """
 from __future__ import annotations
 
-import json
 from typing import Any, Dict, List
 
 from ..api.types import Message, MessageToolCall, ToolCallFunction

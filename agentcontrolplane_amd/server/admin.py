@@ -6,13 +6,11 @@ came from kubectl."""
 from __future__ import annotations
 
 import json
-from typing import Any, Dict
 
 from fastapi import Request
 from fastapi.responses import JSONResponse
 
 from ..api.types import KINDS
-from ..store import AlreadyExistsError
 
 
 def add_admin_routes(app, store) -> None:
