@@ -81,7 +81,13 @@ class Scheduler:
         # loop's next turn always extends the previous one).  KV stays a
         # derived cache — eviction only costs recompute.
         self.retired: Dict[int, tuple] = {}      # seq_id -> token tuple (FIFO)
-        self._retired_index: Dict[tuple, List[int]] = {}
+        # longest-prefix lookup: every retiree is indexed under the rolling
+        # hash of each of its block-aligned prefixes, so an adopter finds the
+        # retiree with the LONGEST shared prefix in O(prompt/block) lookups
+        # (a first-block bucket would keep matching a sibling's short shared
+        # system prompt instead of the same conversation's previous turn)
+        self._prefix_index: Dict[int, List[int]] = {}
+        self._retired_hashes: Dict[int, List[int]] = {}
         self.max_retired = 4096
         self.continuation_cache = True
         self.continuation_hits = 0
@@ -303,31 +309,51 @@ class Scheduler:
 
     # ------------------------------------------------------- continuation
 
-    def _first_block_key(self, tokens) -> tuple:
-        return tuple(tokens[: self.cfg.kv_block_size])
+    def _prefix_hashes(self, tokens, max_blocks: Optional[int] = None) -> List[int]:
+        """Rolling hashes of tokens[:bs], tokens[:2*bs], … (full blocks)."""
+        bs = self.cfg.kv_block_size
+        n_blocks = len(tokens) // bs
+        if max_blocks is not None:
+            n_blocks = min(n_blocks, max_blocks)
+        out: List[int] = []
+        h = 0
+        for b in range(n_blocks):
+            h = hash((h, tuple(tokens[b * bs : (b + 1) * bs])))
+            out.append(h)
+        return out
 
     def _retire(self, seq: Sequence) -> None:
         tokens = tuple(seq.prompt_ids + seq.output_ids)
         cached = self.bm.seq_len(seq.seq_id)
         tokens = tokens[:cached]
+        hashes = self._prefix_hashes(tokens)
+        if not hashes:
+            self.bm.free_seq(seq.seq_id)
+            return
         self.retired[seq.seq_id] = tokens
-        self._retired_index.setdefault(self._first_block_key(tokens), []).append(seq.seq_id)
+        self._retired_hashes[seq.seq_id] = hashes
+        for h in hashes:
+            self._prefix_index.setdefault(h, []).append(seq.seq_id)
         while len(self.retired) > self.max_retired:
             self._evict_one_retired()
+
+    def _unindex(self, old_id: int) -> None:
+        for h in self._retired_hashes.pop(old_id, []):
+            bucket = self._prefix_index.get(h)
+            if bucket is not None:
+                try:
+                    bucket.remove(old_id)
+                except ValueError:
+                    pass
+                if not bucket:
+                    self._prefix_index.pop(h, None)
 
     def _evict_one_retired(self) -> bool:
         if not self.retired:
             return False
         old_id = next(iter(self.retired))
-        tokens = self.retired.pop(old_id)
-        bucket = self._retired_index.get(self._first_block_key(tokens))
-        if bucket is not None:
-            try:
-                bucket.remove(old_id)
-            except ValueError:
-                pass
-            if not bucket:
-                self._retired_index.pop(self._first_block_key(tokens), None)
+        self.retired.pop(old_id)
+        self._unindex(old_id)
         self.bm.free_seq(old_id)
         return True
 
@@ -335,33 +361,29 @@ class Scheduler:
         while self.bm.free_blocks < need_blocks and self._evict_one_retired():
             pass
 
-    #: newest candidates compared per admission (bounds the token compares
-    #: when many conversations share a first block, e.g. a common system
-    #: prompt across 1k tasks)
-    ADOPT_CANDIDATES = 8
-
     def _try_adopt(self, seq: Sequence) -> bool:
-        """Share a retired conversation's KV prefix (refcounted, full blocks
-        only); returns True if the sequence was registered with the block
-        manager.  The retiree stays cached so any number of requests with
-        the same prefix — the next turn of the same task, or sibling tasks
-        sharing a system prompt — reuse it."""
+        """Share the retired conversation with the LONGEST matching KV
+        prefix (refcounted, full blocks only); returns True if the sequence
+        was registered with the block manager.  Retirees stay cached so any
+        number of requests reuse them — the same task's next turn (long
+        match) or sibling tasks sharing a system prompt (short match)."""
         bs = self.cfg.kv_block_size
         if not self.continuation_cache or len(seq.prompt_ids) <= bs:
             return False
-        key = self._first_block_key(seq.prompt_ids)
-        bucket = self._retired_index.get(key, [])
-        best_id, best_len = None, 0
-        for cand in reversed(bucket[-self.ADOPT_CANDIDATES :]):
-            toks = self.retired[cand]
-            limit = min(len(toks), len(seq.prompt_ids) - 1)
-            n = bs  # first block already known equal
-            while n < limit and toks[n] == seq.prompt_ids[n]:
-                n += 1
-            if n > best_len:
-                best_id, best_len = cand, n
-        n_blocks = (best_len // bs) if best_id is not None else 0
-        if n_blocks < 1:
+        # at least one token must remain to prefill (logits come from it)
+        max_blocks = (len(seq.prompt_ids) - 1) // bs
+        hashes = self._prefix_hashes(seq.prompt_ids, max_blocks)
+        best_id, n_blocks = None, 0
+        for b in range(len(hashes) - 1, -1, -1):
+            for cand in self._prefix_index.get(hashes[b], ()):  # longest first
+                toks = self.retired[cand]
+                want = (b + 1) * bs
+                if len(toks) >= want and toks[:want] == tuple(seq.prompt_ids[:want]):
+                    best_id, n_blocks = cand, b + 1
+                    break
+            if best_id is not None:
+                break
+        if best_id is None or n_blocks < 1:
             return False
         self.bm.share_prefix(seq.seq_id, best_id, n_blocks, n_blocks * bs)
         seq.num_processed = n_blocks * bs
