@@ -76,6 +76,7 @@ class ResourceStore:
         wal_path: Optional[str] = None,
         fsync: str = "interval",
         fsync_interval_s: float = 0.05,
+        compact_bytes: int = 256 << 20,
     ):
         self._lock = threading.RLock()
         # kind -> namespace -> name -> obj
@@ -92,6 +93,10 @@ class ResourceStore:
         self._fsync = fsync
         self._fsync_interval = fsync_interval_s
         self._last_fsync = 0.0
+        # auto-compaction: rewrite the log as a snapshot once it exceeds
+        # this size (etcd's compaction role); 0 disables
+        self._compact_bytes = compact_bytes
+        self._wal_written = 0
         if wal_path:
             os.makedirs(os.path.dirname(os.path.abspath(wal_path)), exist_ok=True)
             if os.path.exists(wal_path):
@@ -138,7 +143,13 @@ class ResourceStore:
     def _append_wal(self, op: str, obj: Dict[str, Any]) -> None:
         if self._wal_file is None:
             return
-        self._wal_file.write(json.dumps({"op": op, "obj": obj}, separators=(",", ":")) + "\n")
+        line = json.dumps({"op": op, "obj": obj}, separators=(",", ":")) + "\n"
+        self._wal_file.write(line)
+        self._wal_written += len(line)
+        if self._compact_bytes and self._wal_written > self._compact_bytes:
+            self._wal_written = 0
+            self.compact()
+            return
         self._wal_file.flush()
         if self._fsync == "always":
             os.fsync(self._wal_file.fileno())

@@ -135,3 +135,20 @@ def test_events_recorded(store):
     store.record_event(obj, "Normal", "ValidationSucceeded", "ok")
     evs = store.events_for("t1")
     assert len(evs) == 1 and evs[0]["reason"] == "ValidationSucceeded"
+
+
+def test_wal_auto_compaction(tmp_path):
+    wal = str(tmp_path / "acp.wal")
+    s = ResourceStore(wal_path=wal, compact_bytes=20_000)
+    obj = s.create(make_resource(TASK, "t1"))
+    for i in range(400):
+        obj = s.get(TASK, "t1")
+        obj["status"]["phase"] = f"phase-{i}"
+        obj["status"]["contextWindow"] = [{"role": "user", "content": "x" * 100}]
+        s.update_status(obj)
+    size = os.path.getsize(wal)
+    assert size < 120_000, f"WAL did not auto-compact: {size}"
+    s.close()
+    s2 = ResourceStore(wal_path=wal)
+    assert s2.get(TASK, "t1")["status"]["phase"] == "phase-399"
+    s2.close()
