@@ -462,3 +462,34 @@ def test_sub_agent_delegation(cp):
     # and its answer is the parent's tool result
     tool_msgs = [m for m in task["status"]["contextWindow"] if m["role"] == "tool"]
     assert tool_msgs[0]["content"] == "parent done"  # child used same mock → same answer
+
+
+def test_failing_tool_feeds_back(cp):
+    """An MCP tool error completes the ToolCall with status Error; the loop
+    proceeds (reference checkToolCalls treats Error as completed)."""
+    def boom(**_):
+        raise RuntimeError("tool exploded")
+
+    cp.mcp.register_inproc("bad", {"explode": boom})
+    cp.store.create(make_resource(LLM, "llm1", spec={"provider": "mock"}))
+    cp.store.create(make_resource(MCP_SERVER, "bad", spec={"transport": "inproc"}))
+    cp.store.create(
+        make_resource(
+            AGENT, "a1",
+            spec={"llmRef": {"name": "llm1"}, "system": "s", "mcpServers": [{"name": "bad"}]},
+        )
+    )
+    cp.store.create(
+        make_resource(TASK, "t1", spec={"agentRef": {"name": "a1"}, "userMessage": "x"})
+    )
+    task = wait_for(
+        lambda: (cp.store.get(TASK, "t1") or {}).get("status", {}).get("phase")
+        == TaskPhase.FINAL_ANSWER
+        and cp.store.get(TASK, "t1"),
+        timeout=30,
+    )
+    tcs = cp.store.list(TOOL_CALL, label_selector={"acp.humanlayer.dev/task": "t1"})
+    assert tcs[0]["status"]["status"] == "Error"
+    assert "tool exploded" in tcs[0]["status"]["error"]
+    # the loop still completed with a final answer
+    assert task["status"]["output"] == "mock final answer"
