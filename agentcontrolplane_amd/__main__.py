@@ -85,6 +85,65 @@ def cmd_apply(args) -> None:
                     print(f"{kind.lower()}/{name} error: {r.text}", file=sys.stderr)
 
 
+# printer columns per kind (the reference's kubebuilder printcolumn UX,
+# e.g. task_types.go:195-207)
+_COLUMNS = {
+    "Task": [
+        ("NAME", lambda o: o["metadata"]["name"]),
+        ("AGENT", lambda o: o.get("spec", {}).get("agentRef", {}).get("name", "")),
+        ("PHASE", lambda o: o.get("status", {}).get("phase", "")),
+        ("READY", lambda o: str(o.get("status", {}).get("ready", False))),
+        ("PREVIEW", lambda o: o.get("status", {}).get("userMsgPreview", "")[:40]),
+    ],
+    "Agent": [
+        ("NAME", lambda o: o["metadata"]["name"]),
+        ("READY", lambda o: str(o.get("status", {}).get("ready", False))),
+        ("STATUS", lambda o: o.get("status", {}).get("status", "")),
+        ("DETAIL", lambda o: o.get("status", {}).get("statusDetail", "")[:48]),
+    ],
+    "LLM": [
+        ("NAME", lambda o: o["metadata"]["name"]),
+        ("PROVIDER", lambda o: o.get("spec", {}).get("provider", "")),
+        ("READY", lambda o: str(o.get("status", {}).get("ready", False))),
+        ("DETAIL", lambda o: o.get("status", {}).get("statusDetail", "")[:48]),
+    ],
+    "ToolCall": [
+        ("NAME", lambda o: o["metadata"]["name"]),
+        ("PHASE", lambda o: o.get("status", {}).get("phase", "")),
+        ("TASK", lambda o: o.get("spec", {}).get("taskRef", {}).get("name", "")),
+        ("TOOL", lambda o: o.get("spec", {}).get("toolRef", {}).get("name", "")),
+        ("ERROR", lambda o: o.get("status", {}).get("error", "")[:40]),
+    ],
+    "MCPServer": [
+        ("NAME", lambda o: o["metadata"]["name"]),
+        ("TRANSPORT", lambda o: o.get("spec", {}).get("transport", "")),
+        ("CONNECTED", lambda o: str(o.get("status", {}).get("connected", False))),
+        ("TOOLS", lambda o: str(len(o.get("status", {}).get("tools", []) or []))),
+    ],
+    "ContactChannel": [
+        ("NAME", lambda o: o["metadata"]["name"]),
+        ("TYPE", lambda o: o.get("spec", {}).get("type", "")),
+        ("READY", lambda o: str(o.get("status", {}).get("ready", False))),
+        ("DETAIL", lambda o: o.get("status", {}).get("statusDetail", "")[:48]),
+    ],
+}
+
+
+def _print_table(kind: str, objs) -> None:
+    cols = _COLUMNS.get(kind)
+    if not cols:
+        print(json.dumps(objs, indent=2))
+        return
+    rows = [[str(fn(o)) for _, fn in cols] for o in objs]
+    widths = [
+        max(len(h), *(len(r[i]) for r in rows)) if rows else len(h)
+        for i, (h, _) in enumerate(cols)
+    ]
+    print("  ".join(h.ljust(w) for (h, _), w in zip(cols, widths)))
+    for r in rows:
+        print("  ".join(c.ljust(w) for c, w in zip(r, widths)))
+
+
 def cmd_get(args) -> None:
     import httpx
 
@@ -93,7 +152,15 @@ def cmd_get(args) -> None:
     if args.name:
         url += f"/{args.name}"
     r = client.get(url, params={"namespace": args.namespace})
-    print(json.dumps(r.json(), indent=2))
+    data = r.json()
+    if args.output == "json" or args.name:
+        print(json.dumps(data, indent=2))
+        return
+    from .api.types import KINDS
+
+    canonical = {k.lower(): k for k in KINDS}
+    canonical.update({k.lower() + "s": k for k in KINDS})
+    _print_table(canonical.get(args.kind.lower(), args.kind), data)
 
 
 def main() -> None:
@@ -121,6 +188,7 @@ def main() -> None:
     g.add_argument("name", nargs="?")
     g.add_argument("--namespace", default="default")
     g.add_argument("--server", default="http://127.0.0.1:8082")
+    g.add_argument("-o", "--output", default="table", choices=["table", "json"])
     g.set_defaults(fn=cmd_get)
 
     args = p.parse_args()
