@@ -39,6 +39,15 @@ class FlatBatch:
     logit_rows: torch.Tensor
     sample_seq_ids: List[int]
 
+    # speculative step: decode-row tokens come from the PREVIOUS step's
+    # GPU-resident sampled tokens (gathered by these row indices) instead of
+    # committed CPU values — the engine fills token_ids[num_prefill:] before
+    # the forward launch
+    spec_src_rows: Optional[torch.Tensor] = None
+    #: set by the engine's sample launch; successors gather their decode
+    #: tokens from it
+    _tokens_gpu: Optional[torch.Tensor] = None
+
     # lazily-built device metadata, shared by every layer's kernels
     _prefill_meta: Optional["PrefillMeta"] = None
     _decode_tables_i32: Optional[torch.Tensor] = None

@@ -126,6 +126,10 @@ class EngineConfig:
     max_model_len: int = 8192
     max_queue: int = 65536
     enforce_eager: bool = False          # True disables hipGraph capture
+    # launch decode step t+1 from GPU-resident sampled tokens before step t
+    # syncs (hides per-step host time); auto-disabled for TP and whenever a
+    # grammar-constrained decode is running
+    async_scheduling: bool = True
     tensor_parallel: int = 1
     seed: int = 0
     request_timeout_s: float = 600.0

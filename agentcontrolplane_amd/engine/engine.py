@@ -297,36 +297,67 @@ class InferenceEngine:
     # ------------------------------------------------------------ the loop
 
     def _loop(self) -> None:
+        pending = None  # launched-but-uncommitted step
+        spec_ok = self.cfg.async_scheduling and self.tp_world == 1
         while True:
             with self._work:
-                while self._running and not self._pending and not self.scheduler.has_work():
+                while (
+                    self._running
+                    and not self._pending
+                    and not self.scheduler.has_work()
+                    and pending is None
+                ):
                     self._work.wait(timeout=0.2)
                 if not self._running:
-                    return
-                pending, self._pending = self._pending, []
-            for req in pending:
+                    break
+                new, self._pending = self._pending, []
+            for req in new:
                 self.scheduler.add_request(req)
             try:
                 t0 = time.monotonic()
-                worked = self.step()
+                if pending is None:
+                    pending = self._schedule_and_launch(None)
+                    self._m["busy_time_s"] += time.monotonic() - t0
+                    continue
+                # a step is in flight: launch its successor speculatively,
+                # then sync/commit the pending one
+                nxt = self._schedule_and_launch(pending["batch"]) if spec_ok else None
+                if nxt is not None:
+                    self._m["spec_steps"] = self._m.get("spec_steps", 0) + 1
+                t2 = time.monotonic()
+                self._commit(pending)
+                self._m["sample_time_s"] += time.monotonic() - t2
+                pending = nxt
                 self._m["busy_time_s"] += time.monotonic() - t0
             except Exception as e:  # noqa: BLE001 — fail all in-flight requests
                 import traceback
 
                 traceback.print_exc()
+                pending = None
                 for s in list(self.scheduler.running) + list(self.scheduler.waiting):
                     self.scheduler.abort(s, e)
+        # drain on shutdown so waiters are not left hanging
+        if pending is not None:
+            try:
+                self._commit(pending)
+            except Exception:
+                pass
 
-    def step(self) -> bool:
-        """One scheduler round + forward + sampling.  Returns True if any
-        work ran."""
+    def _schedule_and_launch(self, spec_after):
+        """schedule (optionally speculative) + forward + sample launch;
+        returns the pending-step dict or None."""
         t0 = time.monotonic()
-        out = self.scheduler.schedule()
+        out = self.scheduler.schedule(spec_after=spec_after)
         t1 = time.monotonic()
         self._m["preemptions"] += len(out.preempted)
+        self._m["sched_time_s"] += t1 - t0
         if out.batch is None:
-            return False
+            return None
         batch = out.batch
+        if batch.spec_src_rows is not None and spec_after is not None:
+            # decode-row tokens: gather from the previous step's GPU samples
+            P = batch.num_prefill_tokens
+            batch.token_ids[P:] = spec_after._tokens_gpu[batch.spec_src_rows]
         if self.tp_world > 1 and self.tp_rank == 0:
             from ..engine.batch import batch_to_wire
             from ..parallel.tp import broadcast_step
@@ -339,23 +370,30 @@ class InferenceEngine:
                 self._m["graph_steps"] = self._m.get("graph_steps", 0) + 1
         if logits is None:
             logits = self.model.forward(batch)
-        t2 = time.monotonic()
-        self._sample_and_commit(batch, logits)
-        t3 = time.monotonic()
-        self._m["sched_time_s"] += t1 - t0
-        self._m["compute_time_s"] += t2 - t1   # launch-side; sample syncs
-        self._m["sample_time_s"] += t3 - t2
+        self._m["compute_time_s"] += time.monotonic() - t1
         self._m["steps"] += 1
         self._m["prompt_tokens"] += batch.num_prefill_tokens
+        pending = self._launch_sample(batch, logits)
+        return pending
+
+    def step(self) -> bool:
+        """One serial scheduler round + forward + sampling (tests and the
+        TP worker path).  Returns True if any work ran."""
+        pending = self._schedule_and_launch(None)
+        if pending is None:
+            return False
+        self._commit(pending)
         return True
 
-    def _sample_and_commit(self, batch, logits: torch.Tensor) -> None:
+    def _launch_sample(self, batch, logits: torch.Tensor):
+        """Build sampling inputs and launch the sampler (async on GPU);
+        returns the pending-step record for a later _commit."""
         seqs: List[Sequence] = []
         for sid in batch.sample_seq_ids:
             seqs.append(self.scheduler.seq_by_id(sid))
+        if not seqs:
+            return {"batch": batch, "seqs": [], "tokens": None}
         B = len(seqs)
-        if B == 0:
-            return
         live = logits[:, :N_SPECIAL]  # sampling restricted to decodable ids
         # per-sequence sampling params are static: cache the device tensors
         # against the batch's id tuple (stable across decode steps)
@@ -377,8 +415,7 @@ class InferenceEngine:
         else:
             temps, top_ks, top_ps = cached
         mask = None
-        any_grammar = any(s.grammar is not None for s in seqs)
-        if any_grammar:
+        if any(s.grammar is not None for s in seqs):
             import numpy as np
 
             m = np.ones((B, N_SPECIAL), dtype=bool)
@@ -391,8 +428,18 @@ class InferenceEngine:
                     m[i, list(allowed)] = True
             mask = torch.from_numpy(m).to(logits.device, non_blocking=True)
         tokens = ops.softmax_sample(live, temps, top_ks, top_ps, self._gen, mask)
-        tokens_cpu = tokens.tolist()
+        batch._tokens_gpu = tokens  # speculative successors gather from this
+        return {"batch": batch, "seqs": seqs, "tokens": tokens}
+
+    def _commit(self, pending) -> None:
+        """Sync the sampled tokens and advance sequence/grammar state."""
+        seqs = pending["seqs"]
+        if not seqs:
+            return
+        tokens_cpu = pending["tokens"].tolist()  # the step's one sync point
         for s, tok in zip(seqs, tokens_cpu):
+            if s is None or s.state == "finished":
+                continue  # finished at an earlier commit (speculative row)
             self.scheduler.append_sampled(s, tok)
             self._m["generated_tokens"] += 1
             if s.request.first_token_time is None:
@@ -406,7 +453,7 @@ class InferenceEngine:
                     self.scheduler.finish_seq(s, "tool_calls")
                     continue
                 s.grammar.advance(tok)
-                if len(s.output_ids) >= s.request.sampling.max_tokens * 8:
+                if len(s.request.output_ids) >= s.request.sampling.max_tokens * 8:
                     # runaway guard; grammar's closing mode should prevent this
                     self._m["requests_completed"] += 1
                     self.scheduler.finish_seq(s, "length")
@@ -416,7 +463,9 @@ class InferenceEngine:
                 s.output_ids.pop()
                 self._m["requests_completed"] += 1
                 self.scheduler.finish_seq(s, "stop")
-            elif len(s.output_ids) >= s.request.sampling.max_tokens:
+            elif len(s.request.output_ids) >= s.request.sampling.max_tokens:
+                # request-level count: survives recompute preemption (the
+                # per-sequence list folds into the prompt on preempt)
                 self._m["requests_completed"] += 1
                 self.scheduler.finish_seq(s, "length")
 

@@ -96,6 +96,19 @@ class PyBlockManager:
         self._lens[seq_id] = start + n_tokens
         return slots
 
+    def pop_last_token(self, seq_id: int) -> None:
+        """Undo the last append_tokens(seq, 1) — the speculative-step
+        rollback when the previous step finished the sequence."""
+        cur = self._lens[seq_id]
+        assert cur > 0
+        self._lens[seq_id] = cur - 1
+        table = self._tables[seq_id]
+        if (cur - 1) % self.block_size == 0 and len(table) * self.block_size >= cur:
+            # the popped token was alone in the trailing block
+            b = table.pop()
+            self._release_block(b)
+            self.table_epoch += 1
+
     def adopt_prefix(self, new_seq_id: int, old_seq_id: int, n_blocks: int,
                      n_tokens: int) -> None:
         """Transfer the first n_blocks of old's table to a NEW sequence (the

@@ -109,8 +109,16 @@ class Scheduler:
 
     # ---------------------------------------------------------------- step
 
-    def schedule(self) -> SchedulerOutput:
-        """Build the next step's FlatBatch."""
+    def schedule(self, spec_after: Optional[FlatBatch] = None) -> SchedulerOutput:
+        """Build the next step's FlatBatch.
+
+        With ``spec_after`` (the still-uncommitted previous batch), build a
+        SPECULATIVE step: decode rows take their input token from the
+        previous step's GPU-resident samples (``spec_src_rows``), so the
+        step can launch before the previous one syncs.  Speculation bails
+        (batch=None) rather than preempt, and only when every running
+        decode sequence is speculable (no grammar-constrained decodes —
+        their masks depend on the uncommitted token)."""
         preempted: List[Sequence] = []
         bs = self.cfg.kv_block_size
 
@@ -123,13 +131,33 @@ class Scheduler:
 
         # 1) reserve one decode token per running decode sequence; preempt
         #    the youngest until the reservations fit the pool
-        decode_seqs = [s for s in self.running if s.state == DECODE]
+        spec_src: Dict[int, int] = {}
+        if spec_after is not None:
+            for i, sid in enumerate(spec_after.sample_seq_ids):
+                spec_src[sid] = i
+            decode_seqs = []
+            for s in self.running:
+                if s.state != DECODE:
+                    continue
+                if s.grammar is not None or s.seq_id not in spec_src:
+                    return SchedulerOutput(batch=None, preempted=preempted)
+                if len(s.request.output_ids) + 2 > s.request.sampling.max_tokens:
+                    continue  # finishes by length at the pending commit
+                decode_seqs.append(s)
+            if not decode_seqs and not self.waiting and not any(
+                s.state == PREFILL for s in self.running
+            ):
+                return SchedulerOutput(batch=None, preempted=preempted)
+        else:
+            decode_seqs = [s for s in self.running if s.state == DECODE]
         while decode_seqs:
             decode_need = sum(seq_blocks_for(s, 1) for s in decode_seqs)
             if decode_need <= self.bm.free_blocks:
                 break
             if self._evict_one_retired():
                 continue  # reclaimed cache blocks before preempting live work
+            if spec_after is not None:
+                return SchedulerOutput(batch=None, preempted=preempted)
             victim = decode_seqs.pop()  # youngest
             self.running.remove(victim)
             self._preempt(victim)
@@ -197,7 +225,8 @@ class Scheduler:
             budget -= chunk
             avail -= blocks_for(cur, have, chunk)
 
-        decode_seqs = [s for s in self.running if s.state == DECODE]
+        if spec_after is None:
+            decode_seqs = [s for s in self.running if s.state == DECODE]
         if not prefills and not decode_seqs:
             return SchedulerOutput(batch=None, preempted=preempted)
 
@@ -239,11 +268,17 @@ class Scheduler:
         decode_seq_lens = None
         decode_tables_i32 = None
         decode_ids: List[int] = []
+        spec_rows: List[int] = []
         if decode_seqs:
             lens = []
             for s in decode_seqs:
-                tok = s.output_ids[-1] if s.output_ids else s.prompt_ids[-1]
-                pos = s.total_len - 1
+                if spec_after is not None:
+                    tok = 0  # filled on device from the previous step's samples
+                    pos = s.total_len  # the in-flight token occupies total_len
+                    spec_rows.append(spec_src[s.seq_id])
+                else:
+                    tok = s.output_ids[-1] if s.output_ids else s.prompt_ids[-1]
+                    pos = s.total_len - 1
                 slots = self.bm.append_tokens(s.seq_id, 1)
                 token_ids.append(tok)
                 positions.append(pos)
@@ -284,6 +319,10 @@ class Scheduler:
         )
         if decode_tables_i32 is not None:
             batch._decode_tables_i32 = decode_tables_i32
+        if spec_after is not None and spec_rows:
+            batch.spec_src_rows = torch.tensor(
+                spec_rows, device=self.device, dtype=torch.long
+            )
         return SchedulerOutput(batch=batch, preempted=preempted)
 
     # ------------------------------------------------------------ commit

@@ -68,6 +68,21 @@ class BlockManager {
     return slots;
   }
 
+  void pop_last_token(int64_t seq_id) {
+    auto lit = lens_.find(seq_id);
+    if (lit == lens_.end() || lit->second <= 0)
+      throw std::runtime_error("pop_last_token: empty seq");
+    const int64_t cur = lit->second;
+    lit->second = cur - 1;
+    auto& table = tables_.at(seq_id);
+    if ((cur - 1) % block_size_ == 0 &&
+        (int64_t)table.size() * block_size_ >= cur) {
+      release_block(table.back());
+      table.pop_back();
+      ++table_epoch_;
+    }
+  }
+
   void adopt_prefix(int64_t new_seq_id, int64_t old_seq_id, int64_t n_blocks,
                     int64_t n_tokens) {
     if (tables_.count(new_seq_id))
@@ -163,6 +178,7 @@ void register_block_manager(py::module_& m) {
       .def("can_append", &BlockManager::can_append)
       .def("append_tokens", &BlockManager::append_tokens)
       .def("adopt_prefix", &BlockManager::adopt_prefix)
+      .def("pop_last_token", &BlockManager::pop_last_token)
       .def("share_prefix", &BlockManager::share_prefix)
       .def("ref_count", &BlockManager::ref_count)
       .def("block_table", &BlockManager::block_table)
