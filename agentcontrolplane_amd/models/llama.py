@@ -165,7 +165,7 @@ class LlamaForCausalLM:
                 normed = ops.rmsnorm(x, lw.input_norm, cfg.rms_eps)
             else:
                 normed, residual = ops.fused_add_rmsnorm(x, residual, lw.input_norm, cfg.rms_eps)
-            qkv = F.linear(normed, lw.qkv)
+            qkv = ops.linear(normed, lw.qkv)
             # strided views into the fused GEMM output: the HIP kernels take
             # a token stride, so no .contiguous() copies on the hot path
             row = qkv.stride(0)
@@ -181,15 +181,15 @@ class LlamaForCausalLM:
                 self.k_caches[li], self.v_caches[li], self.cos_sin,
             )
             attn = self._attention(li, q, batch)
-            o = F.linear(attn.reshape(N, qd), lw.o)
+            o = ops.linear(attn.reshape(N, qd), lw.o)
             if self.all_reduce is not None:
                 o = self.all_reduce(o)
             normed2, residual = ops.fused_add_rmsnorm(o, residual, lw.post_norm, cfg.rms_eps)
-            gate_up = F.linear(normed2, lw.gate_up)
-            mlp = F.linear(ops.swiglu(gate_up), lw.down)
+            gate_up = ops.linear(normed2, lw.gate_up)
+            mlp = ops.linear(ops.swiglu(gate_up), lw.down)
             if self.all_reduce is not None:
                 mlp = self.all_reduce(mlp)
             x = mlp
         normed, _ = ops.fused_add_rmsnorm(x, residual, self.final_norm, cfg.rms_eps)
         sel = normed[batch.logit_rows]
-        return F.linear(sel, self.lm_head).float()
+        return ops.linear(sel, self.lm_head).float()

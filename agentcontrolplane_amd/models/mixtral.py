@@ -83,8 +83,8 @@ class MixtralForCausalLM(LlamaForCausalLM):
                 continue
             rows = flat_rows[sel]
             xe = x[rows]
-            gate_up = F.linear(xe, self.expert_gate_up[li][e])
-            ye = F.linear(ops.swiglu(gate_up), self.expert_down[li][e])
+            gate_up = ops.linear(xe, self.expert_gate_up[li][e])
+            ye = ops.linear(ops.swiglu(gate_up), self.expert_down[li][e])
             out.index_add_(0, rows, ye * flat_w[sel].unsqueeze(-1))
         return out
 
@@ -117,7 +117,7 @@ class MixtralForCausalLM(LlamaForCausalLM):
                 normed = ops.rmsnorm(x, lw.input_norm, cfg.rms_eps)
             else:
                 normed, residual = ops.fused_add_rmsnorm(x, residual, lw.input_norm, cfg.rms_eps)
-            qkv = F.linear(normed, lw.qkv)
+            qkv = ops.linear(normed, lw.qkv)
             row = qkv.stride(0)
             q = qkv.as_strided((N, self.n_heads, self.head_dim), (row, self.head_dim, 1))
             k = qkv.as_strided(
@@ -131,7 +131,7 @@ class MixtralForCausalLM(LlamaForCausalLM):
                 self.k_caches[li], self.v_caches[li], self.cos_sin,
             )
             attn = self._attention(li, q, batch)
-            o = F.linear(attn.reshape(N, qd), lw.o)
+            o = ops.linear(attn.reshape(N, qd), lw.o)
             if self.all_reduce is not None:
                 o = self.all_reduce(o)
             normed2, residual = ops.fused_add_rmsnorm(o, residual, lw.post_norm, cfg.rms_eps)
@@ -142,4 +142,4 @@ class MixtralForCausalLM(LlamaForCausalLM):
                 x = self.all_reduce(x)
         normed, _ = ops.fused_add_rmsnorm(x, residual, self.final_norm, cfg.rms_eps)
         sel = normed[batch.logit_rows]
-        return F.linear(sel, self.lm_head).float()
+        return ops.linear(sel, self.lm_head).float()

@@ -42,6 +42,29 @@ def _impl(t: torch.Tensor):
     return m
 
 
+import os
+
+# below this M the library GEMM (hipBLASLt skinny kernels) wins; above it the
+# hand-written 256²-tile MFMA GEMM does (A/B in tools/gemm_bench.py)
+_GEMM_MIN_M = int(os.environ.get("ACP_GEMM_MIN_M", "1024"))
+
+
+def linear(x, w):
+    """x[*, K] @ w[N, K]^T → [*, N].  Large-M bf16 shapes go to the
+    hand-written 8-phase MFMA GEMM (csrc/gemm_bf16.hip); everything else
+    (decode-sized M, ragged shards, CPU) to torch.nn.functional.linear."""
+    if (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and x.dim() == 2
+        and x.shape[0] >= _GEMM_MIN_M
+        and w.shape[0] % 256 == 0
+        and w.shape[1] % 128 == 0
+    ):
+        return _hip().gemm_bf16(x, w)
+    return torch.nn.functional.linear(x, w)
+
+
 def rmsnorm(x, weight, eps: float = 1e-5):
     return _impl(x).rmsnorm(x, weight, eps)
 

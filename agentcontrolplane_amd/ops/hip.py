@@ -93,6 +93,12 @@ def attention_decode_raw(q, k_cache, v_cache, block_tables, seq_lens, scale):
     return out
 
 
+def gemm_bf16(x, w):
+    out = torch.empty((x.shape[0], w.shape[0]), dtype=x.dtype, device=x.device)
+    _C.gemm_bf16(out, x.contiguous(), w)
+    return out
+
+
 def swiglu(gate_up):
     inter = gate_up.shape[-1] // 2
     out = torch.empty(

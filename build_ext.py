@@ -30,6 +30,7 @@ def build(verbose: bool = False) -> str:
         os.path.join(CSRC, "prefill_attn.hip"),
         os.path.join(CSRC, "prefill_attn_mfma.hip"),
         os.path.join(CSRC, "mfma_probe.hip"),
+        os.path.join(CSRC, "gemm_bf16.hip"),
         os.path.join(CSRC, "sampling.hip"),
     ]
     module = cpp_extension.load(

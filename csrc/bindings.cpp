@@ -30,6 +30,7 @@ void launch_prefill_attn_mfma(void*, const void*, const void*, const void*, cons
                               const int*, const int*, const int*, const int*, const int*,
                               float, int, int, int, int, int, int, int64_t, hipStream_t);
 void launch_mfma_probe(float*, const void*, const void*, hipStream_t);
+void launch_gemm_bf16(void*, const void*, const void*, int, int, int, hipStream_t);
 }
 
 #define CHECK_CUDA(x) TORCH_CHECK(x.is_cuda(), #x " must be a GPU tensor")
@@ -161,6 +162,20 @@ static void prefill_attn_mfma(torch::Tensor out, torch::Tensor q, torch::Tensor 
                     row_starts, tile_seq, tile_q0, scale, /*mfma=*/true);
 }
 
+// C[M,N] = A[M,K] @ B[N,K]^T — the hand-written 256²-tile 8-phase MFMA
+// GEMM (csrc/gemm_bf16.hip) for large-M bf16 projections
+static void gemm_bf16(torch::Tensor C, torch::Tensor A, torch::Tensor B) {
+  CHECK_CUDA(A); CHECK_CONT(A); CHECK_BF16(A);
+  CHECK_CONT(B); CHECK_BF16(B); CHECK_CONT(C); CHECK_BF16(C);
+  const int M = A.size(0), K = A.size(1), N = B.size(0);
+  TORCH_CHECK(B.size(1) == K && C.size(0) == M && C.size(1) == N,
+              "gemm_bf16 shape mismatch");
+  TORCH_CHECK(K % 128 == 0, "gemm_bf16 requires K % 128 == 0");
+  TORCH_CHECK(N % 256 == 0, "gemm_bf16 requires N % 256 == 0");
+  launch_gemm_bf16(C.data_ptr(), A.data_ptr(), B.data_ptr(), M, N, K,
+                   cur_stream());
+}
+
 static torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
   CHECK_CUDA(a); CHECK_BF16(a); CHECK_CONT(a); CHECK_CONT(b);
   auto d = torch::empty({32, 32}, a.options().dtype(at::kFloat));
@@ -197,5 +212,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("prefill_attn", &prefill_attn);
   m.def("prefill_attn_mfma", &prefill_attn_mfma);
   m.def("mfma_probe", &mfma_probe);
+  m.def("gemm_bf16", &gemm_bf16);
   m.def("sample", &sample);
 }

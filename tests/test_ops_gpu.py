@@ -264,3 +264,25 @@ def test_model_forward_gpu_vs_cpu_reference():
     scale = clog.abs().max().item()
     err = (glog - clog).abs().max().item()
     assert err / max(scale, 1e-6) < 0.08, f"rel err {err / scale} (scale {scale})"
+
+
+@requires_gpu
+@pytest.mark.parametrize(
+    "m,n,k",
+    [
+        (256, 256, 128),      # single tile, one 8-phase iteration
+        (300, 512, 256),      # ragged M (row-clamp + store guard)
+        (1024, 1536, 4096),   # multi-tile, deep K
+        (2048, 768, 1664),    # K % 128 == 0 but K % 256 != 0; odd tile grid
+    ],
+)
+def test_gemm_bf16(m, n, k):
+    """Hand-written 256²-tile 8-phase MFMA GEMM vs fp32 matmul oracle."""
+    torch.manual_seed(11)
+    x = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
+    got = hip().gemm_bf16(x, w)
+    want = x.float() @ w.float().t()
+    # bf16 inputs, fp32 accumulation: error scales with sqrt(K)
+    tol = 0.03 * math.sqrt(k)  # bf16 output rounding of ~4.5σ results
+    assert maxerr(got, want) < tol, maxerr(got, want)
