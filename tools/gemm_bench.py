@@ -10,7 +10,11 @@ chunked-prefill M regime (budget 8192) plus decode-sized M for the
 threshold choice, and the 70B TP=8 shards.
 """
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 
