@@ -73,8 +73,20 @@ __global__ __launch_bounds__(512, 1) void gemm_bf16_kernel(
   const int q = nwg >> 3, r = nwg & 7;
   const int xcd = blockIdx.x & 7, idx = blockIdx.x >> 3;
   const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
-  const int tm0 = (wg % tiles_m) * BM;
-  const int tn0 = (wg / tiles_m) * BN;
+  // 8(m)x4(n) supertile rasterization: concurrent blocks share 8 A-tiles +
+  // 4 B-tiles (k-slab fronts stay L2-resident) instead of sweeping all of
+  // M per N-column, which re-streams A from HBM tiles_n times (ragged
+  // edges handled by clamped band/column widths; bijective)
+  const int tiles_n = nwg / tiles_m;
+  const int per_sc = tiles_m * 4;          // tiles per 4-wide super-column
+  const int sc = wg / per_sc;
+  const int rsc = wg % per_sc;
+  const int h = min(4, tiles_n - sc * 4);  // n-width of this super-column
+  const int band = 8 * h;                  // tiles per 8-high m-band
+  const int b = rsc / band, rb = rsc % band;
+  const int w = min(8, tiles_m - b * 8);   // m-height of this band
+  const int tm0 = (b * 8 + rb % w) * BM;
+  const int tn0 = (sc * 4 + rb / w) * BN;
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
