@@ -52,6 +52,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=50)
     ap.add_argument("--check", action="store_true")
+    ap.add_argument("--only", default=None, help="substring filter on shape tag")
     args = ap.parse_args()
 
     from agentcontrolplane_amd.ops import hip
@@ -60,6 +61,8 @@ def main():
     print(f"{'shape':14s} {'M':>5s} {'N':>6s} {'K':>6s}  {'lib µs':>8s} {'own µs':>8s}"
           f"  {'lib TF':>7s} {'own TF':>7s}  win")
     for tag, m, n, k in SHAPES:
+        if args.only and args.only not in tag:
+            continue
         x = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
         w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
         fl = 2.0 * m * n * k
