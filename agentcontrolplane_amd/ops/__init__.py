@@ -44,9 +44,12 @@ def _impl(t: torch.Tensor):
 
 import os
 
-# below this M the library GEMM (hipBLASLt skinny kernels) wins; above it the
-# hand-written 256²-tile MFMA GEMM does (A/B in tools/gemm_bench.py)
-_GEMM_MIN_M = int(os.environ.get("ACP_GEMM_MIN_M", "1024"))
+# Measured A/B (tools/gemm_bench.py, profiles/r01_gemm_ab.md): the TunableOp-
+# tuned hipBLASLt stream-K kernels hold 1.3-1.6 PF on every engine projection
+# shape; the hand-written 8-phase kernel (csrc/gemm_bf16.hip) reaches 1.1 PF
+# (zero LDS conflicts, barrier/issue-stall bound).  The library stays the
+# default; set ACP_GEMM_MIN_M to route M >= threshold to the custom kernel.
+_GEMM_MIN_M = int(os.environ.get("ACP_GEMM_MIN_M", str(1 << 30)))
 
 
 def linear(x, w):
