@@ -373,7 +373,11 @@ class InferenceEngine:
         self._m["compute_time_s"] += time.monotonic() - t1
         self._m["steps"] += 1
         self._m["prompt_tokens"] += batch.num_prefill_tokens
+        t2 = time.monotonic()
         pending = self._launch_sample(batch, logits)
+        self._m["sample_launch_time_s"] = (
+            self._m.get("sample_launch_time_s", 0.0) + time.monotonic() - t2
+        )
         return pending
 
     def step(self) -> bool:
@@ -418,6 +422,7 @@ class InferenceEngine:
         if any(s.grammar is not None for s in seqs):
             import numpy as np
 
+            tm = time.monotonic()
             m = np.ones((B, N_SPECIAL), dtype=bool)
             for i, s in enumerate(seqs):
                 if s.grammar is not None:
@@ -427,6 +432,9 @@ class InferenceEngine:
                     m[i] = False
                     m[i, list(allowed)] = True
             mask = torch.from_numpy(m).to(logits.device, non_blocking=True)
+            self._m["mask_time_s"] = (
+                self._m.get("mask_time_s", 0.0) + time.monotonic() - tm
+            )
         tokens = ops.softmax_sample(live, temps, top_ks, top_ps, self._gen, mask)
         batch._tokens_gpu = tokens  # speculative successors gather from this
         return {"batch": batch, "seqs": seqs, "tokens": tokens}

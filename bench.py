@@ -345,6 +345,10 @@ def main() -> None:
                     "engine_sched_s": round(em["sched_time_s"], 2),
                     "engine_compute_s": round(em["compute_time_s"], 2),
                     "engine_sample_s": round(em["sample_time_s"], 2),
+                    "engine_sample_launch_s": round(em.get("sample_launch_time_s", 0.0), 2),
+                    "engine_mask_s": round(em.get("mask_time_s", 0.0), 2),
+                    "engine_spec_steps": em.get("spec_steps", 0),
+                    "engine_graph_steps": em.get("graph_steps", 0),
                     "kv_occupancy": round(em["kv_occupancy"], 4),
                     "preemptions": em["preemptions"],
                 },
