@@ -392,6 +392,7 @@ class InferenceEngine:
     def _launch_sample(self, batch, logits: torch.Tensor):
         """Build sampling inputs and launch the sampler (async on GPU);
         returns the pending-step record for a later _commit."""
+        ta = time.monotonic()
         seqs: List[Sequence] = []
         for sid in batch.sample_seq_ids:
             seqs.append(self.scheduler.seq_by_id(sid))
@@ -399,6 +400,8 @@ class InferenceEngine:
             return {"batch": batch, "seqs": [], "tokens": None}
         B = len(seqs)
         live = logits[:, :N_SPECIAL]  # sampling restricted to decodable ids
+        self._m["ls_gather_s"] = self._m.get("ls_gather_s", 0.0) + time.monotonic() - ta
+        tb = time.monotonic()
         # per-sequence sampling params are static: cache the device tensors
         # against the batch's id tuple (stable across decode steps)
         key = tuple(batch.sample_seq_ids)
@@ -435,7 +438,10 @@ class InferenceEngine:
             self._m["mask_time_s"] = (
                 self._m.get("mask_time_s", 0.0) + time.monotonic() - tm
             )
+        self._m["ls_params_s"] = self._m.get("ls_params_s", 0.0) + time.monotonic() - tb
+        tc = time.monotonic()
         tokens = ops.softmax_sample(live, temps, top_ks, top_ps, self._gen, mask)
+        self._m["ls_kernel_s"] = self._m.get("ls_kernel_s", 0.0) + time.monotonic() - tc
         batch._tokens_gpu = tokens  # speculative successors gather from this
         return {"batch": batch, "seqs": seqs, "tokens": tokens}
 

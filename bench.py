@@ -347,6 +347,7 @@ def main() -> None:
                     "engine_sample_s": round(em["sample_time_s"], 2),
                     "engine_sample_launch_s": round(em.get("sample_launch_time_s", 0.0), 2),
                     "engine_mask_s": round(em.get("mask_time_s", 0.0), 2),
+                    "engine_ls": [round(em.get(k, 0.0), 2) for k in ("ls_gather_s", "ls_params_s", "ls_kernel_s")],
                     "engine_spec_steps": em.get("spec_steps", 0),
                     "engine_graph_steps": em.get("graph_steps", 0),
                     "kv_occupancy": round(em["kv_occupancy"], 4),
