@@ -50,6 +50,9 @@ class FlatBatch:
 
     # lazily-built device metadata, shared by every layer's kernels
     _prefill_meta: Optional["PrefillMeta"] = None
+    #: pinned-staging helper (engine/stage.py); set by the scheduler so the
+    #: per-step metadata uploads are async instead of blocking hipMemcpys
+    _stager: Optional[object] = None
     _decode_tables_i32: Optional[torch.Tensor] = None
     _decode_lens_i32: Optional[torch.Tensor] = None
 
@@ -66,7 +69,7 @@ class FlatBatch:
             self._prefill_meta = {}
         if tile_q not in self._prefill_meta:
             self._prefill_meta[tile_q] = PrefillMeta.build(
-                self.prefills, self.token_ids.device, tile_q
+                self.prefills, self.token_ids.device, tile_q, self._stager
             )
         return self._prefill_meta[tile_q]
 
@@ -143,7 +146,8 @@ class PrefillMeta:
     tile_q: int
 
     @staticmethod
-    def build(prefills: List[SeqMeta], device, tile_q: int = 64) -> "PrefillMeta":
+    def build(prefills: List[SeqMeta], device, tile_q: int = 64,
+              stager=None) -> "PrefillMeta":
         max_blocks = max((len(m.block_table) for m in prefills), default=1)
         tables, seq_lens, ctx_lens, row_starts = [], [], [], []
         tile_seq, tile_q0 = [], []
@@ -157,13 +161,16 @@ class PrefillMeta:
                 tile_seq.append(i)
                 tile_q0.append(q0)
             row += m.query_len
-        t = lambda x: torch.tensor(x, device=device, dtype=torch.int32)  # noqa: E731
+        if stager is not None:
+            t = lambda n, x: stager.tensor(f"pm{tile_q}_{n}", x, "int32")  # noqa: E731
+        else:
+            t = lambda n, x: torch.tensor(x, device=device, dtype=torch.int32)  # noqa: E731
         return PrefillMeta(
-            block_tables=t(tables),
-            seq_lens=t(seq_lens),
-            ctx_lens=t(ctx_lens),
-            row_starts=t(row_starts),
-            tile_seq=t(tile_seq),
-            tile_q0=t(tile_q0),
+            block_tables=t("bt", tables),
+            seq_lens=t("sl", seq_lens),
+            ctx_lens=t("cl", ctx_lens),
+            row_starts=t("rs", row_starts),
+            tile_seq=t("ts", tile_seq),
+            tile_q0=t("tq", tile_q0),
             tile_q=tile_q,
         )

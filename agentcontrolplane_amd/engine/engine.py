@@ -407,16 +407,13 @@ class InferenceEngine:
         key = tuple(batch.sample_seq_ids)
         cached = self._sampling_cache if self._sampling_cache_key == key else None
         if cached is None:
-            temps = torch.tensor(
-                [s.request.sampling.temperature for s in seqs], device=logits.device
-            )
-            top_ks = torch.tensor(
-                [s.request.sampling.top_k for s in seqs], device=logits.device,
-                dtype=torch.long,
-            )
-            top_ps = torch.tensor(
-                [s.request.sampling.top_p for s in seqs], device=logits.device
-            )
+            # fresh device tensors staged through pinned memory: a plain
+            # torch.tensor(..., device="cuda") is a BLOCKING copy that
+            # drains the speculative pipeline (profiles/r01: 5.2 s/bench)
+            stager = self.scheduler.stager
+            temps = stager.fresh([s.request.sampling.temperature for s in seqs], "float32")
+            top_ks = stager.fresh([s.request.sampling.top_k for s in seqs], "int64")
+            top_ps = stager.fresh([s.request.sampling.top_p for s in seqs], "float32")
             self._sampling_cache_key = key
             self._sampling_cache = (temps, top_ks, top_ps)
         else:
@@ -434,7 +431,7 @@ class InferenceEngine:
                         allowed = set(allowed) | {EOT}
                     m[i] = False
                     m[i, list(allowed)] = True
-            mask = torch.from_numpy(m).to(logits.device, non_blocking=True)
+            mask = self.scheduler.stager.tensor("mask", m, "bool")
             self._m["mask_time_s"] = (
                 self._m.get("mask_time_s", 0.0) + time.monotonic() - tm
             )

@@ -64,9 +64,12 @@ class SchedulerOutput:
 
 class Scheduler:
     def __init__(self, config: EngineConfig, block_manager, device):
+        from .stage import HostStager
+
         self.cfg = config
         self.bm = block_manager
         self.device = torch.device(device)
+        self.stager = HostStager(device)
         self.waiting: List[Sequence] = []
         self.running: List[Sequence] = []   # in admission order (oldest first)
         self._by_id: Dict[int, Sequence] = {}
@@ -120,6 +123,7 @@ class Scheduler:
         decode sequence is speculable (no grammar-constrained decodes —
         their masks depend on the uncommitted token)."""
         preempted: List[Sequence] = []
+        self.stager.step()  # rotate the pinned staging slots (stage.py)
         bs = self.cfg.kv_block_size
 
         def blocks_for(cur_len: int, have_blocks: int, n_tokens: int) -> int:
@@ -294,35 +298,34 @@ class Scheduler:
             if key == self._decode_cache_key:
                 decode_block_tables, decode_tables_i32 = self._decode_cache
             else:
+                import numpy as np
+
                 tables = [self.bm.block_table(i) for i in decode_ids]
                 max_blocks = max(len(t) for t in tables)
                 padded = [t + [0] * (max_blocks - len(t)) for t in tables]
-                decode_block_tables = torch.tensor(
-                    padded, device=self.device, dtype=torch.long
-                )
+                decode_block_tables = self.stager.fresh(padded, np.int64)
                 decode_tables_i32 = decode_block_tables.int()
                 self._decode_cache_key = key
                 self._decode_cache = (decode_block_tables, decode_tables_i32)
-            decode_seq_lens = torch.tensor(lens, device=self.device, dtype=torch.long)
+            decode_seq_lens = self.stager.tensor("dlens", lens, "int64")
 
         batch = FlatBatch(
-            token_ids=torch.tensor(token_ids, device=self.device, dtype=torch.long),
-            positions=torch.tensor(positions, device=self.device, dtype=torch.long),
-            slot_mapping=torch.tensor(slot_mapping, device=self.device, dtype=torch.long),
+            token_ids=self.stager.tensor("tok", token_ids, "int64"),
+            positions=self.stager.tensor("pos", positions, "int64"),
+            slot_mapping=self.stager.tensor("slot", slot_mapping, "int64"),
             prefills=prefill_metas,
             num_prefill_tokens=num_prefill_tokens,
             decode_seq_ids=decode_ids,
             decode_block_tables=decode_block_tables,
             decode_seq_lens=decode_seq_lens,
-            logit_rows=torch.tensor(logit_rows, device=self.device, dtype=torch.long),
+            logit_rows=self.stager.tensor("lrow", logit_rows, "int64"),
             sample_seq_ids=sample_seq_ids,
         )
+        batch._stager = self.stager
         if decode_tables_i32 is not None:
             batch._decode_tables_i32 = decode_tables_i32
         if spec_after is not None and spec_rows:
-            batch.spec_src_rows = torch.tensor(
-                spec_rows, device=self.device, dtype=torch.long
-            )
+            batch.spec_src_rows = self.stager.tensor("spec", spec_rows, "int64")
         return SchedulerOutput(batch=batch, preempted=preempted)
 
     # ------------------------------------------------------------ commit
