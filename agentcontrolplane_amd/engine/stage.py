@@ -13,10 +13,16 @@ before the slot is reused at t+3; stream order protects the device side).
 """
 from __future__ import annotations
 
+import os
 from typing import Dict, Tuple
 
 import numpy as np
 import torch
+
+# staging scope: "full" = all per-step metadata, "params" = only the
+# cross-step cached tensors (sampling params, decode tables), "off" =
+# plain blocking torch.tensor everywhere (the pre-stager behavior)
+_MODE = os.environ.get("ACP_STAGE_MODE", "full")
 
 _TORCH_DTYPE = {
     np.dtype(np.int64): torch.int64,
@@ -57,6 +63,8 @@ class HostStager:
         tdt = _TORCH_DTYPE[arr.dtype]
         if not self.cuda:
             return torch.from_numpy(arr.copy())
+        if _MODE != "full":
+            return torch.from_numpy(arr).to(self.device, non_blocking=True)
         n = arr.size
         buf = self._stage_pinned((name, self._slot), arr, tdt)
         dev = self._dev.get(name)
@@ -73,6 +81,8 @@ class HostStager:
         tdt = _TORCH_DTYPE[arr.dtype]
         if not self.cuda:
             return torch.from_numpy(arr.copy())
+        if _MODE == "off":
+            return torch.from_numpy(arr).to(self.device)
         self._fresh_n += 1
         buf = self._stage_pinned(("__fresh__", self._fresh_n, self._slot), arr, tdt)
         dev = torch.empty(arr.shape, dtype=tdt, device=self.device)
