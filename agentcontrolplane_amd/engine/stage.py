@@ -19,10 +19,14 @@ from typing import Dict, Tuple
 import numpy as np
 import torch
 
-# staging scope: "full" = all per-step metadata, "params" = only the
-# cross-step cached tensors (sampling params, decode tables), "off" =
-# plain blocking torch.tensor everywhere (the pre-stager behavior)
-_MODE = os.environ.get("ACP_STAGE_MODE", "full")
+# staging scope (A/B'd on MI355X, profiles/r01_gemm_ab.md methodology):
+#   "params" (default) — pinned+async only for the cross-step cached tensors
+#       (sampling params, decode tables); per-step metadata goes
+#       np.asarray → .to(device, non_blocking).  85.7 steps/s.
+#   "full" — pinned rotating buffers for every per-step tensor: the extra
+#       per-op staging cost outweighs the drain it avoids (60.4 steps/s).
+#   "off" — blocking copies everywhere (84.7 steps/s).
+_MODE = os.environ.get("ACP_STAGE_MODE", "params")
 
 _TORCH_DTYPE = {
     np.dtype(np.int64): torch.int64,
