@@ -19,14 +19,18 @@ from typing import Dict, Tuple
 import numpy as np
 import torch
 
-# staging scope (A/B'd on MI355X, profiles/r01_gemm_ab.md methodology):
-#   "params" (default) — pinned+async only for the cross-step cached tensors
-#       (sampling params, decode tables); per-step metadata goes
-#       np.asarray → .to(device, non_blocking).  85.7 steps/s.
-#   "full" — pinned rotating buffers for every per-step tensor: the extra
-#       per-op staging cost outweighs the drain it avoids (60.4 steps/s).
-#   "off" — blocking copies everywhere (84.7 steps/s).
-_MODE = os.environ.get("ACP_STAGE_MODE", "params")
+# staging scope, A/B'd on MI355X (default bench @128 tasks / @1k tasks):
+#   "numpy" (default) — np.asarray → .to(device) everywhere.  The original
+#       torch.tensor(python_list, device="cuda") cost 5.2 s/bench in list
+#       parsing + blocking copies; the numpy route alone recovers it
+#       (84.7 steps/s @128, 66.7 @1k).
+#   "params" — additionally stages cross-step cached tensors (sampling
+#       params, decode tables) through pinned memory: ties @128 (85.7) but
+#       collapses @1k (40.8) — MB-scale pinned re-allocs for the 1000-row
+#       decode tables are device-synchronizing.
+#   "full" — pinned rotating buffers for every per-step tensor: per-op
+#       staging cost outweighs the drains it avoids (60.4 @128).
+_MODE = os.environ.get("ACP_STAGE_MODE", "numpy")
 
 _TORCH_DTYPE = {
     np.dtype(np.int64): torch.int64,
@@ -85,7 +89,7 @@ class HostStager:
         tdt = _TORCH_DTYPE[arr.dtype]
         if not self.cuda:
             return torch.from_numpy(arr.copy())
-        if _MODE == "off":
+        if _MODE != "full" and _MODE != "params":
             return torch.from_numpy(arr).to(self.device)
         self._fresh_n += 1
         buf = self._stage_pinned(("__fresh__", self._fresh_n, self._slot), arr, tdt)
