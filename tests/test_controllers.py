@@ -493,3 +493,41 @@ def test_failing_tool_feeds_back(cp):
     assert "tool exploded" in tcs[0]["status"]["error"]
     # the loop still completed with a final answer
     assert task["status"]["output"] == "mock final answer"
+
+
+def test_task_delete_mid_loop_cascades(cp):
+    """Deleting a Task while ToolCalls are pending cascade-deletes the
+    children (owner index) and leaves the manager healthy for new work."""
+    make_basic_world(cp)
+    cp.store.create(
+        make_resource(TASK, "tdel", spec={"agentRef": {"name": "a1"}, "userMessage": "add 1 2"})
+    )
+    # wait until the ToolCall child exists, then delete the parent
+    tc = wait_for(
+        lambda: next(
+            iter(
+                cp.store.list(TOOL_CALL, label_selector={"acp.humanlayer.dev/task": "tdel"})
+            ),
+            None,
+        ),
+        timeout=20,
+    )
+    cp.store.delete(TASK, "tdel")
+    wait_for(
+        lambda: not cp.store.list(
+            TOOL_CALL, label_selector={"acp.humanlayer.dev/task": "tdel"}
+        ),
+        timeout=10,
+    )
+    assert cp.store.get(TASK, "tdel") is None
+    # the control plane still runs new tasks to completion afterwards
+    cp.store.create(
+        make_resource(TASK, "tafter", spec={"agentRef": {"name": "a1"}, "userMessage": "add 2 3"})
+    )
+    task = wait_for(
+        lambda: (cp.store.get(TASK, "tafter") or {}).get("status", {}).get("phase")
+        == TaskPhase.FINAL_ANSWER
+        and cp.store.get(TASK, "tafter"),
+        timeout=20,
+    )
+    assert task["status"]["output"] == "mock final answer"

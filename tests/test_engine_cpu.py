@@ -186,3 +186,22 @@ def test_kv_cache_consistency_incremental_vs_full(engine):
         assert r1.output_ids == r2.output_ids
     finally:
         eng2.stop()
+
+
+def test_request_timeout_surfaces():
+    """A request that cannot finish inside request_timeout_s raises
+    TimeoutError from generate() (LocalEngineClient maps it to a retryable
+    503, llmclient/local.py)."""
+    cfg = EngineConfig(
+        model="tiny", device="cpu", num_kv_blocks=256, kv_block_size=4,
+        max_prefill_tokens=8, request_timeout_s=0.15,
+    )
+    eng = InferenceEngine(cfg)
+    try:
+        with pytest.raises(TimeoutError):
+            eng.generate(
+                list(range(200)),  # 25 prefill steps at budget 8
+                SamplingParams(max_tokens=64, temperature=0),
+            )
+    finally:
+        eng.stop()
