@@ -327,6 +327,16 @@ class InferenceEngine:
                 t2 = time.monotonic()
                 self._commit(pending)
                 self._m["sample_time_s"] += time.monotonic() - t2
+                if nxt is not None and nxt.get("deferred"):
+                    # grammar-carrying speculative step: its forward is
+                    # already on the GPU; launch the sampler now that the
+                    # commit above advanced the grammar states its masks
+                    # depend on
+                    t3 = time.monotonic()
+                    nxt = self._launch_sample(nxt["batch"], nxt["logits"])
+                    self._m["sample_launch_time_s"] = (
+                        self._m.get("sample_launch_time_s", 0.0) + time.monotonic() - t3
+                    )
                 pending = nxt
                 self._m["busy_time_s"] += time.monotonic() - t0
             except Exception as e:  # noqa: BLE001 — fail all in-flight requests
@@ -339,6 +349,8 @@ class InferenceEngine:
         # drain on shutdown so waiters are not left hanging
         if pending is not None:
             try:
+                if pending.get("deferred"):
+                    pending = self._launch_sample(pending["batch"], pending["logits"])
                 self._commit(pending)
             except Exception:
                 pass
@@ -373,6 +385,16 @@ class InferenceEngine:
         self._m["compute_time_s"] += time.monotonic() - t1
         self._m["steps"] += 1
         self._m["prompt_tokens"] += batch.num_prefill_tokens
+        if spec_after is not None and any(
+            self.scheduler.seq_by_id(sid) is not None
+            and self.scheduler.seq_by_id(sid).grammar is not None
+            for sid in batch.sample_seq_ids
+        ):
+            # masks depend on the uncommitted predecessor token: defer the
+            # sampler launch to after the predecessor's commit (the loop
+            # resolves this before scheduling OUR successor, so
+            # _tokens_gpu exists by the time it is gathered from)
+            return {"batch": batch, "logits": logits, "deferred": True}
         t2 = time.monotonic()
         pending = self._launch_sample(batch, logits)
         self._m["sample_launch_time_s"] = (
@@ -410,22 +432,26 @@ class InferenceEngine:
             # fresh device tensors staged through pinned memory: a plain
             # torch.tensor(..., device="cuda") is a BLOCKING copy that
             # drains the speculative pipeline (profiles/r01: 5.2 s/bench)
+            # a row's sequence may have finished at the commit that just
+            # ran (deferred grammar speculation): sample it with defaults,
+            # the commit discards its token
             stager = self.scheduler.stager
-            temps = stager.fresh([s.request.sampling.temperature for s in seqs], "float32")
-            top_ks = stager.fresh([s.request.sampling.top_k for s in seqs], "int64")
-            top_ps = stager.fresh([s.request.sampling.top_p for s in seqs], "float32")
+            sp = [s.request.sampling if s is not None else None for s in seqs]
+            temps = stager.fresh([p.temperature if p else 0.0 for p in sp], "float32")
+            top_ks = stager.fresh([p.top_k if p else 0 for p in sp], "int64")
+            top_ps = stager.fresh([p.top_p if p else 1.0 for p in sp], "float32")
             self._sampling_cache_key = key
             self._sampling_cache = (temps, top_ks, top_ps)
         else:
             temps, top_ks, top_ps = cached
         mask = None
-        if any(s.grammar is not None for s in seqs):
+        if any(s is not None and s.grammar is not None for s in seqs):
             import numpy as np
 
             tm = time.monotonic()
             m = np.ones((B, N_SPECIAL), dtype=bool)
             for i, s in enumerate(seqs):
-                if s.grammar is not None:
+                if s is not None and s.grammar is not None:
                     allowed = s.grammar.allowed_tokens()
                     if s.grammar.accepting:
                         allowed = set(allowed) | {EOT}

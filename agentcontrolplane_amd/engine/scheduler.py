@@ -143,9 +143,15 @@ class Scheduler:
             for s in self.running:
                 if s.state != DECODE:
                     continue
-                if s.grammar is not None or s.seq_id not in spec_src:
+                if s.seq_id not in spec_src:
                     return SchedulerOutput(batch=None, preempted=preempted)
-                if len(s.request.output_ids) + 2 > s.request.sampling.max_tokens:
+                # grammar-constrained rows ARE speculable: the forward does
+                # not depend on the pending token, only the sampler's mask
+                # does — the engine defers the sampler launch of a
+                # grammar-carrying speculative step until the predecessor's
+                # commit has advanced the grammar states.
+                cap = s.request.sampling.max_tokens * (8 if s.grammar is not None else 1)
+                if len(s.request.output_ids) + 2 > cap:
                     continue  # finishes by length at the pending commit
                 decode_seqs.append(s)
             if not decode_seqs and not self.waiting and not any(

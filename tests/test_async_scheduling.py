@@ -62,8 +62,11 @@ def test_speculation_with_eot_finishers():
     assert want == got
 
 
-def test_speculation_disabled_with_constrained():
-    """Grammar-constrained decodes force serial steps; results stay valid."""
+def test_speculation_with_constrained_decodes():
+    """Grammar-constrained decodes stay in the pipeline (the sampler launch
+    of a grammar-carrying speculative step is deferred until the
+    predecessor's commit advances the PDA states); results stay valid and
+    speculative steps actually happen."""
     import json
 
     eng = make_engine(True)
@@ -78,6 +81,7 @@ def test_speculation_disabled_with_constrained():
         )
         assert res.finish_reason == "tool_calls"
         json.loads(res.tool_calls[0]["function"]["arguments"])
+        assert eng.metrics().get("spec_steps", 0) > 0
     finally:
         eng.stop()
 
