@@ -53,6 +53,9 @@ class FlatBatch:
     #: pinned-staging helper (engine/stage.py); set by the scheduler so the
     #: per-step metadata uploads are async instead of blocking hipMemcpys
     _stager: Optional[object] = None
+    #: host-side lists mirroring the device tensors (set by the scheduler)
+    #: so batch_to_wire never syncs
+    _host: Optional[dict] = None
     _decode_tables_i32: Optional[torch.Tensor] = None
     _decode_lens_i32: Optional[torch.Tensor] = None
 
@@ -86,24 +89,28 @@ class FlatBatch:
 
 def batch_to_wire(batch: FlatBatch) -> dict:
     """Compact picklable form for the TP metadata broadcast (rank 0 runs the
-    scheduler; other ranks rebuild the batch and run the sharded forward)."""
+    scheduler; other ranks rebuild the batch and run the sharded forward).
+    Prefers the scheduler's host-side lists (``_host``) so the broadcast
+    never D2H-syncs the just-uploaded device tensors."""
+    h = getattr(batch, "_host", None) or {}
+
+    def of(name, tensor):
+        v = h.get(name)
+        return v if v is not None else (tensor.tolist() if tensor is not None else None)
+
     return {
-        "token_ids": batch.token_ids.tolist(),
-        "positions": batch.positions.tolist(),
-        "slot_mapping": batch.slot_mapping.tolist(),
+        "token_ids": of("token_ids", batch.token_ids),
+        "positions": of("positions", batch.positions),
+        "slot_mapping": of("slot_mapping", batch.slot_mapping),
         "prefills": [
             (m.seq_id, m.query_len, m.seq_len, m.ctx_len, m.block_table, m.needs_logits)
             for m in batch.prefills
         ],
         "num_prefill_tokens": batch.num_prefill_tokens,
         "decode_seq_ids": batch.decode_seq_ids,
-        "decode_block_tables": (
-            batch.decode_block_tables.tolist() if batch.decode_block_tables is not None else None
-        ),
-        "decode_seq_lens": (
-            batch.decode_seq_lens.tolist() if batch.decode_seq_lens is not None else None
-        ),
-        "logit_rows": batch.logit_rows.tolist(),
+        "decode_block_tables": of("decode_block_tables", batch.decode_block_tables),
+        "decode_seq_lens": of("decode_seq_lens", batch.decode_seq_lens),
+        "logit_rows": of("logit_rows", batch.logit_rows),
         "sample_seq_ids": batch.sample_seq_ids,
     }
 

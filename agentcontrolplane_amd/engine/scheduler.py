@@ -119,9 +119,9 @@ class Scheduler:
         SPECULATIVE step: decode rows take their input token from the
         previous step's GPU-resident samples (``spec_src_rows``), so the
         step can launch before the previous one syncs.  Speculation bails
-        (batch=None) rather than preempt, and only when every running
-        decode sequence is speculable (no grammar-constrained decodes —
-        their masks depend on the uncommitted token)."""
+        (batch=None) rather than preempt, and requires every running decode
+        sequence to appear in the pending step's sample rows (grammar rows
+        are included — the engine defers only their sampler launch)."""
         preempted: List[Sequence] = []
         self.stager.step()  # rotate the pinned staging slots (stage.py)
         bs = self.cfg.kv_block_size
@@ -313,6 +313,7 @@ class Scheduler:
                 decode_tables_i32 = decode_block_tables.int()
                 self._decode_cache_key = key
                 self._decode_cache = (decode_block_tables, decode_tables_i32)
+                self._decode_cache_host = padded
             decode_seq_lens = self.stager.tensor("dlens", lens, "int64")
 
         batch = FlatBatch(
@@ -328,6 +329,17 @@ class Scheduler:
             sample_seq_ids=sample_seq_ids,
         )
         batch._stager = self.stager
+        # host-side copies for the TP wire broadcast (batch_to_wire):
+        # avoids per-step D2H tolist() syncs on rank 0
+        batch._host = {
+            "token_ids": token_ids,
+            "positions": positions,
+            "slot_mapping": slot_mapping,
+            "logit_rows": logit_rows,
+            "decode_block_tables": getattr(self, "_decode_cache_host", None)
+            if decode_block_tables is not None else None,
+            "decode_seq_lens": lens if decode_seqs else None,
+        }
         if decode_tables_i32 is not None:
             batch._decode_tables_i32 = decode_tables_i32
         if spec_after is not None and spec_rows:
