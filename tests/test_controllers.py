@@ -531,3 +531,47 @@ def test_task_delete_mid_loop_cascades(cp):
         timeout=20,
     )
     assert task["status"]["output"] == "mock final answer"
+
+
+def test_soak_concurrent_task_mix(cp):
+    """Soak: 40 concurrent tasks against one plane — plain tool loops plus
+    mid-flight deletions racing the reconcilers.  Everything must reach a
+    terminal phase (or be cleanly gone) with no stuck reconciles."""
+    make_basic_world(cp)
+    n = 40
+    for i in range(n):
+        cp.store.create(
+            make_resource(
+                TASK, f"soak-{i}",
+                spec={"agentRef": {"name": "a1"}, "userMessage": f"add {i} 1"},
+            )
+        )
+    # delete a few while they run
+    import time as _t
+
+    _t.sleep(0.2)
+    for i in range(0, n, 10):
+        try:
+            cp.store.delete(TASK, f"soak-{i}")
+        except Exception:
+            pass
+
+    def all_done():
+        for i in range(n):
+            t = cp.store.get(TASK, f"soak-{i}")
+            if t is None:
+                continue
+            if t.get("status", {}).get("phase") not in (
+                TaskPhase.FINAL_ANSWER, TaskPhase.FAILED,
+            ):
+                return None
+        return True
+
+    assert wait_for(all_done, timeout=60)
+    # the survivors all carry the full checkpoint
+    done = [cp.store.get(TASK, f"soak-{i}") for i in range(n)]
+    done = [t for t in done if t is not None]
+    assert len(done) >= n - 4
+    for t in done:
+        assert t["status"]["output"] == "mock final answer"
+        assert [m["role"] for m in t["status"]["contextWindow"]][-1] == "assistant"
