@@ -269,6 +269,49 @@ class InferenceEngine:
         req.wait(self.cfg.request_timeout_s)
         return self._result_of(req, t0)
 
+    def chat_stream(
+        self,
+        messages: List[Dict[str, Any]],
+        tools: Optional[List[Dict[str, Any]]] = None,
+        sampling: Optional[SamplingParams] = None,
+    ):
+        """Streaming chat: yields ("delta", text) per decoded UTF-8 span as
+        tokens commit, then exactly one ("done", ChatResult).  Byte tokens
+        are buffered until they form valid UTF-8 (multi-byte codepoints
+        split across tokens).  Constrained (tool-call) turns yield no
+        deltas — the tool_calls arrive on the final result."""
+        import queue as _queue
+
+        t0 = time.monotonic()
+        req = self._build_request(messages, tools, sampling)
+        q: "_queue.Queue" = _queue.Queue()
+        if not req.constrained:
+            req.on_token = q.put
+        req.on_complete = lambda r: q.put(None)
+        self.submit(req)
+        buf = bytearray()
+        deadline = t0 + self.cfg.request_timeout_s
+        while True:
+            try:
+                tok = q.get(timeout=max(0.0, deadline - time.monotonic()))
+            except _queue.Empty:
+                raise TimeoutError(f"request {req.request_id} timed out") from None
+            if tok is None:
+                break
+            if tok < 256:
+                buf.append(tok)
+                try:
+                    text = buf.decode("utf-8")
+                except UnicodeDecodeError:
+                    continue  # mid-codepoint: wait for the next byte
+                buf.clear()
+                yield ("delta", text)
+        if req.error is not None:
+            raise req.error
+        if buf:
+            yield ("delta", buf.decode("utf-8", errors="replace"))
+        yield ("done", self._result_of(req, t0))
+
     def chat_async(self, messages, tools, sampling, callback) -> InferenceRequest:
         """Submit a chat turn; ``callback(result, error)`` fires from the
         engine thread on completion.  This is the serving path: reconciler
@@ -500,7 +543,10 @@ class InferenceEngine:
                 s.output_ids.pop()
                 self._m["requests_completed"] += 1
                 self.scheduler.finish_seq(s, "stop")
-            elif len(s.request.output_ids) >= s.request.sampling.max_tokens:
+                continue
+            if s.request.on_token is not None:
+                s.request.on_token(tok)
+            if len(s.request.output_ids) >= s.request.sampling.max_tokens:
                 # request-level count: survives recompute preemption (the
                 # per-sequence list folds into the prompt on preempt)
                 self._m["requests_completed"] += 1

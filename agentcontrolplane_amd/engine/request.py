@@ -57,6 +57,9 @@ class InferenceRequest:
         # invoked (from the engine thread) when the request completes; used
         # by the async serving path to requeue the owning Task reconcile
         self.on_complete = None
+        # invoked (from the engine thread) per committed free-text token —
+        # the streaming path; must be cheap and non-blocking (queue.put)
+        self.on_token = None
         # scheduler state
         self.seq = None  # assigned by the scheduler
 

@@ -365,6 +365,45 @@ def build_app(store, manager=None, engine=None) -> FastAPI:
             top_p=float(body.get("top_p", 1.0)),
             tool_choice=tool_choice if isinstance(tool_choice, str) else "required",
         )
+        import time as _time
+        import uuid as _uuid
+
+        if body.get("stream"):
+            # OpenAI SSE chunks; the sync generator runs in Starlette's
+            # threadpool and pulls per-token deltas off the engine thread
+            from fastapi.responses import StreamingResponse
+
+            cid = f"chatcmpl-{_uuid.uuid4().hex[:24]}"
+            created = int(_time.time())
+            model_name = body.get("model", engine.cfg.model)
+
+            def chunk(delta, finish=None):
+                return "data: " + json.dumps({
+                    "id": cid, "object": "chat.completion.chunk",
+                    "created": created, "model": model_name,
+                    "choices": [{"index": 0, "delta": delta, "finish_reason": finish}],
+                }) + "\n\n"
+
+            def sse():
+                yield chunk({"role": "assistant"})
+                try:
+                    for kind, payload in engine.chat_stream(messages, tools, sampling):
+                        if kind == "delta":
+                            yield chunk({"content": payload})
+                        else:
+                            finish = payload.finish_reason
+                            if payload.tool_calls:
+                                yield chunk({"tool_calls": payload.tool_calls}, None)
+                                finish = "tool_calls"
+                            elif finish not in ("stop", "length"):
+                                finish = "stop"
+                            yield chunk({}, finish)
+                except Exception as e:  # noqa: BLE001 — surfaced as an SSE error event
+                    yield "data: " + json.dumps({"error": {"message": str(e)}}) + "\n\n"
+                yield "data: [DONE]\n\n"
+
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
         try:
             import anyio
 
@@ -386,9 +425,6 @@ def build_app(store, manager=None, engine=None) -> FastAPI:
             msg["tool_calls"] = result.tool_calls
             msg["content"] = None
             finish = "tool_calls"
-        import time as _time
-        import uuid as _uuid
-
         return {
             "id": f"chatcmpl-{_uuid.uuid4().hex[:24]}",
             "object": "chat.completion",

@@ -205,3 +205,19 @@ def test_request_timeout_surfaces():
             )
     finally:
         eng.stop()
+
+
+def test_chat_stream_matches_blocking(engine):
+    """Greedy streaming deltas concatenate to exactly the blocking result."""
+    msgs = [{"role": "user", "content": "stream me"}]
+    want = engine.chat(msgs, sampling=SamplingParams(max_tokens=12, temperature=0))
+    parts, done = [], None
+    for kind, payload in engine.chat_stream(
+        msgs, sampling=SamplingParams(max_tokens=12, temperature=0)
+    ):
+        if kind == "delta":
+            parts.append(payload)
+        else:
+            done = payload
+    assert done is not None and done.finish_reason == want.finish_reason
+    assert "".join(parts) == want.text == done.text
