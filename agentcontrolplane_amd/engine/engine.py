@@ -429,14 +429,13 @@ class InferenceEngine:
         self._m["steps"] += 1
         self._m["prompt_tokens"] += batch.num_prefill_tokens
         if spec_after is not None and any(
-            self.scheduler.seq_by_id(sid) is not None
-            and self.scheduler.seq_by_id(sid).grammar is not None
-            for sid in batch.sample_seq_ids
+            self._stateful_sampling(sid) for sid in batch.sample_seq_ids
         ):
-            # masks depend on the uncommitted predecessor token: defer the
-            # sampler launch to after the predecessor's commit (the loop
-            # resolves this before scheduling OUR successor, so
-            # _tokens_gpu exists by the time it is gathered from)
+            # grammar masks, repetition penalties and per-request seeded
+            # streams all depend on the COMMITTED output: defer the sampler
+            # launch to after the predecessor's commit (the loop resolves
+            # this before scheduling OUR successor, so _tokens_gpu exists
+            # by the time it is gathered from)
             return {"batch": batch, "logits": logits, "deferred": True}
         t2 = time.monotonic()
         pending = self._launch_sample(batch, logits)
@@ -453,6 +452,19 @@ class InferenceEngine:
             return False
         self._commit(pending)
         return True
+
+    def _stateful_sampling(self, sid) -> bool:
+        s = self.scheduler.seq_by_id(sid)
+        if s is None:
+            return False
+        if s.grammar is not None:
+            return True
+        sp = s.request.sampling
+        return (
+            sp.seed is not None
+            or bool(sp.frequency_penalty)
+            or bool(sp.presence_penalty)
+        )
 
     def _launch_sample(self, batch, logits: torch.Tensor):
         """Build sampling inputs and launch the sampler (async on GPU);
@@ -506,7 +518,47 @@ class InferenceEngine:
             )
         self._m["ls_params_s"] = self._m.get("ls_params_s", 0.0) + time.monotonic() - tb
         tc = time.monotonic()
-        tokens = ops.softmax_sample(live, temps, top_ks, top_ps, self._gen, mask)
+        # frequency/presence penalties (llm_types.go BaseConfig): subtract
+        # freq·count + presence·[seen] over the request's committed output
+        if any(
+            s is not None and s.state != "finished"
+            and (s.request.sampling.frequency_penalty or s.request.sampling.presence_penalty)
+            for s in seqs
+        ):
+            import numpy as np
+
+            pen = np.zeros((B, N_SPECIAL), dtype=np.float32)
+            for i, s in enumerate(seqs):
+                if s is None or s.state == "finished":
+                    continue
+                sp2 = s.request.sampling
+                if not (sp2.frequency_penalty or sp2.presence_penalty):
+                    continue
+                counts: Dict[int, int] = {}
+                for t in s.request.output_ids:
+                    if t < N_SPECIAL:
+                        counts[t] = counts.get(t, 0) + 1
+                for t, c in counts.items():
+                    pen[i, t] = sp2.frequency_penalty * c + sp2.presence_penalty
+            live = live - self.scheduler.stager.tensor("pen", pen, "float32")
+        # per-request seeded draws: each seeded request consumes exactly one
+        # uniform from its own generator per committed token
+        uniforms = None
+        if any(
+            s is not None and s.state != "finished"
+            and s.request.sampling.seed is not None
+            for s in seqs
+        ):
+            uniforms = torch.rand(B, device=logits.device, generator=self._gen)
+            for i, s in enumerate(seqs):
+                if s is None or s.state == "finished" or s.request.sampling.seed is None:
+                    continue
+                if s.request.gen is None:
+                    g = torch.Generator(device=logits.device)
+                    g.manual_seed(int(s.request.sampling.seed))
+                    s.request.gen = g
+                uniforms[i] = torch.rand(1, device=logits.device, generator=s.request.gen)[0]
+        tokens = ops.softmax_sample(live, temps, top_ks, top_ps, self._gen, mask, uniforms)
         self._m["ls_kernel_s"] = self._m.get("ls_kernel_s", 0.0) + time.monotonic() - tc
         batch._tokens_gpu = tokens  # speculative successors gather from this
         return {"batch": batch, "seqs": seqs, "tokens": tokens}

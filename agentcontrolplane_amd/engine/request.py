@@ -60,6 +60,9 @@ class InferenceRequest:
         # invoked (from the engine thread) per committed free-text token —
         # the streaming path; must be cheap and non-blocking (queue.put)
         self.on_token = None
+        # per-request RNG for SamplingParams.seed (created on first use by
+        # the engine, on its device)
+        self.gen = None
         # scheduler state
         self.seq = None  # assigned by the scheduler
 

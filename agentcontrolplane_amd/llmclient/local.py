@@ -28,6 +28,8 @@ class LocalEngineClient(LLMClient):
         self.temperature = float(params.get("temperature") or 0.7)
         self.top_p = float(params.get("topP") or 1.0)
         self.top_k = int(params.get("topK") or 0)
+        self.frequency_penalty = float(params.get("frequencyPenalty") or 0.0)
+        self.presence_penalty = float(params.get("presencePenalty") or 0.0)
 
     def _sampling(self):
         from ..engine.request import SamplingParams
@@ -37,6 +39,8 @@ class LocalEngineClient(LLMClient):
             temperature=self.temperature,
             top_p=self.top_p,
             top_k=self.top_k,
+            frequency_penalty=self.frequency_penalty,
+            presence_penalty=self.presence_penalty,
         )
 
     @staticmethod

@@ -106,7 +106,10 @@ def swiglu(gate_up):
     return _impl(gate_up).swiglu(gate_up)
 
 
-def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None):
+def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None, uniforms=None):
     """Batch sampling: temperature/top-k/top-p with optional per-row boolean
-    vocab masks (constrained decoding).  Returns [B] long."""
-    return _impl(logits).softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask)
+    vocab masks (constrained decoding) and optional precomputed per-row
+    uniforms (per-request seeded generators).  Returns [B] long."""
+    return _impl(logits).softmax_sample(
+        logits, temperatures, top_ks, top_ps, gen, mask, uniforms
+    )

@@ -108,10 +108,11 @@ def swiglu(gate_up):
     return out
 
 
-def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None):
+def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None, uniforms=None):
     B, V = logits.shape
     out = torch.empty(B, dtype=torch.long, device=logits.device)
-    uniforms = torch.rand(B, device=logits.device, generator=gen)
+    if uniforms is None:
+        uniforms = torch.rand(B, device=logits.device, generator=gen)
     _C.sample(
         out, logits.float().contiguous(), temperatures.float().contiguous(),
         top_ks.long().contiguous(), top_ps.float().contiguous(),

@@ -150,9 +150,10 @@ def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
     return (torch.nn.functional.silu(gate) * up).to(gate_up.dtype)
 
 
-def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None):
+def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None, uniforms=None):
     """logits: [B, V] float; temperatures/top_ps: [B] float; top_ks: [B] long
-    (0 = off); mask: [B, V] bool (True = allowed) or None.
+    (0 = off); mask: [B, V] bool (True = allowed) or None; uniforms: [B]
+    precomputed draws (per-request seeds) — inverse-CDF like the HIP kernel.
     Greedy when temperature == 0."""
     logits = logits.float().clone()
     B, V = logits.shape
@@ -179,5 +180,9 @@ def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None):
             row = torch.full_like(row, float("-inf"))
             row[sorted_idx] = sorted_logits
         probs = torch.softmax(row, dim=-1)
-        out[i] = torch.multinomial(probs, 1, generator=gen).squeeze(-1)
+        if uniforms is not None:
+            cdf = torch.cumsum(probs, dim=-1)
+            out[i] = int(torch.searchsorted(cdf, uniforms[i].to(cdf) * cdf[-1]).clamp(max=probs.numel() - 1))
+        else:
+            out[i] = torch.multinomial(probs, 1, generator=gen).squeeze(-1)
     return out

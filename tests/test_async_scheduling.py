@@ -170,3 +170,24 @@ def test_stress_mixed_grammar_pressure_equivalence():
         if j["tool"]:
             assert a[i] and a[i][0][0] == "srv__go"
             json.loads(a[i][0][1])
+
+
+def test_seeded_and_penalized_equivalence():
+    """Seeded and penalty-carrying requests are in the deferred-sampler
+    class: outputs identical with the pipeline on and off."""
+    def run(async_sched):
+        eng = make_engine(async_sched)
+        try:
+            outs = []
+            for i in range(6):
+                prompt = [(i * 13 + k) % 250 for k in range(30 + i)]
+                outs.append(eng.generate(
+                    prompt,
+                    SamplingParams(max_tokens=14, temperature=1.0, seed=100 + i,
+                                   frequency_penalty=0.5, presence_penalty=0.2),
+                ).output_ids)
+            return outs
+        finally:
+            eng.stop()
+
+    assert run(False) == run(True)

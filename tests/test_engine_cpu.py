@@ -221,3 +221,25 @@ def test_chat_stream_matches_blocking(engine):
             done = payload
     assert done is not None and done.finish_reason == want.finish_reason
     assert "".join(parts) == want.text == done.text
+
+
+def test_seeded_sampling_reproducible(engine):
+    tok = engine.tokenizer
+    ids = tok.render_chat([{"role": "user", "content": "seeded"}])
+    a = engine.generate(ids, SamplingParams(max_tokens=12, temperature=1.0, seed=42))
+    b = engine.generate(ids, SamplingParams(max_tokens=12, temperature=1.0, seed=42))
+    c = engine.generate(ids, SamplingParams(max_tokens=12, temperature=1.0, seed=43))
+    assert a.output_ids == b.output_ids
+    assert a.output_ids != c.output_ids  # astronomically unlikely to collide
+
+
+def test_frequency_penalty_blocks_repeats(engine):
+    """frequency_penalty large enough dominates any logit gap: every
+    committed token must be unique (vocab 261 >> max_tokens)."""
+    tok = engine.tokenizer
+    ids = tok.render_chat([{"role": "user", "content": "no repeats"}])
+    r = engine.generate(
+        ids,
+        SamplingParams(max_tokens=24, temperature=0.0, frequency_penalty=100.0),
+    )
+    assert len(r.output_ids) == len(set(r.output_ids)), r.output_ids
