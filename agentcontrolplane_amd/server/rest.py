@@ -363,6 +363,10 @@ def build_app(store, manager=None, engine=None) -> FastAPI:
             max_tokens=int(body.get("max_tokens") or 256),
             temperature=float(body.get("temperature", 0.7)),
             top_p=float(body.get("top_p", 1.0)),
+            top_k=int(body.get("top_k", 0)),
+            frequency_penalty=float(body.get("frequency_penalty", 0.0)),
+            presence_penalty=float(body.get("presence_penalty", 0.0)),
+            seed=body.get("seed"),
             tool_choice=tool_choice if isinstance(tool_choice, str) else "required",
         )
         import time as _time
