@@ -103,6 +103,7 @@ class InferenceEngine:
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
         self._pending: List[InferenceRequest] = []
+        self._cancelled: List[InferenceRequest] = []
         self._running = False
         self._thread: Optional[threading.Thread] = None
         # metrics
@@ -205,6 +206,13 @@ class InferenceEngine:
             self._work.notify_all()
         return request
 
+    def cancel(self, req: InferenceRequest) -> None:
+        """Stop generating for ``req`` at the next step boundary (used when
+        a streaming client disconnects).  Safe from any thread."""
+        with self._work:
+            self._cancelled.append(req)
+            self._work.notify_all()
+
     def generate(
         self,
         prompt_ids: List[int],
@@ -291,21 +299,28 @@ class InferenceEngine:
         self.submit(req)
         buf = bytearray()
         deadline = t0 + self.cfg.request_timeout_s
-        while True:
-            try:
-                tok = q.get(timeout=max(0.0, deadline - time.monotonic()))
-            except _queue.Empty:
-                raise TimeoutError(f"request {req.request_id} timed out") from None
-            if tok is None:
-                break
-            if tok < 256:
-                buf.append(tok)
+        try:
+            while True:
                 try:
-                    text = buf.decode("utf-8")
-                except UnicodeDecodeError:
-                    continue  # mid-codepoint: wait for the next byte
-                buf.clear()
-                yield ("delta", text)
+                    tok = q.get(timeout=max(0.0, deadline - time.monotonic()))
+                except _queue.Empty:
+                    raise TimeoutError(f"request {req.request_id} timed out") from None
+                if tok is None:
+                    break
+                if tok < 256:
+                    buf.append(tok)
+                    try:
+                        text = buf.decode("utf-8")
+                    except UnicodeDecodeError:
+                        continue  # mid-codepoint: wait for the next byte
+                    buf.clear()
+                    yield ("delta", text)
+        finally:
+            # the consumer may abandon the stream (client disconnect →
+            # GeneratorExit): stop generating instead of running to
+            # max_tokens
+            if not req._event.is_set():
+                self.cancel(req)
         if req.error is not None:
             raise req.error
         if buf:
@@ -354,8 +369,11 @@ class InferenceEngine:
                 if not self._running:
                     break
                 new, self._pending = self._pending, []
+                cancels, self._cancelled = self._cancelled, []
             for req in new:
                 self.scheduler.add_request(req)
+            for req in cancels:
+                self.scheduler.cancel_request(req)
             try:
                 t0 = time.monotonic()
                 if pending is None:

@@ -463,6 +463,18 @@ class Scheduler:
         seq.state = WAITING
         self.waiting.insert(0, seq)
 
+    def cancel_request(self, request) -> None:
+        """Stop generating for a request (e.g. a streaming client went
+        away).  Runs in the engine loop thread; a finished/unknown request
+        is a no-op.  The sequence finishes with reason "cancelled" — its KV
+        retires into the continuation cache like any other finish."""
+        seq = request.seq
+        if seq is None or seq.state == FINISHED:
+            return
+        if seq in self.waiting:
+            self.waiting.remove(seq)
+        self.finish_seq(seq, "cancelled")
+
     def abort(self, seq: Sequence, err: BaseException) -> None:
         seq.state = FINISHED
         if seq in self.running:

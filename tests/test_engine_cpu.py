@@ -243,3 +243,28 @@ def test_frequency_penalty_blocks_repeats(engine):
         SamplingParams(max_tokens=24, temperature=0.0, frequency_penalty=100.0),
     )
     assert len(r.output_ids) == len(set(r.output_ids)), r.output_ids
+
+
+def test_stream_close_cancels_request(engine):
+    """Abandoning a stream (client disconnect) stops generation instead of
+    running to max_tokens; the engine keeps serving."""
+    import time as _t
+
+    msgs = [{"role": "user", "content": "long stream"}]
+    gen = engine.chat_stream(msgs, sampling=SamplingParams(max_tokens=2000, temperature=1.0))
+    got = 0
+    for kind, _ in gen:
+        if kind == "delta":
+            got += 1
+        if got >= 2:
+            break
+    gen.close()  # GeneratorExit → engine.cancel(req)
+    deadline = _t.monotonic() + 5
+    while _t.monotonic() < deadline:
+        if not engine.scheduler.running and not engine.scheduler.waiting:
+            break
+        _t.sleep(0.02)
+    assert not engine.scheduler.running, "cancelled request still generating"
+    # engine still healthy
+    r = engine.chat(msgs, sampling=SamplingParams(max_tokens=4, temperature=0))
+    assert r.finish_reason in ("stop", "length")
