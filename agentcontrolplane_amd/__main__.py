@@ -163,6 +163,19 @@ def cmd_get(args) -> None:
     _print_table(canonical.get(args.kind.lower(), args.kind), data)
 
 
+def cmd_delete(args) -> None:
+    import httpx
+
+    client = httpx.Client(base_url=args.server, timeout=30)
+    r = client.delete(
+        f"/admin/resources/{args.kind}/{args.name}", params={"namespace": args.namespace}
+    )
+    if r.status_code >= 400:
+        print(r.text, file=sys.stderr)
+        sys.exit(1)
+    print(r.json().get("deleted", ""))
+
+
 def main() -> None:
     p = argparse.ArgumentParser(prog="agentcontrolplane_amd")
     sub = p.add_subparsers(dest="cmd", required=True)
@@ -190,6 +203,13 @@ def main() -> None:
     g.add_argument("--server", default="http://127.0.0.1:8082")
     g.add_argument("-o", "--output", default="table", choices=["table", "json"])
     g.set_defaults(fn=cmd_get)
+
+    d = sub.add_parser("delete", help="delete a resource (cascades to owned children)")
+    d.add_argument("kind")
+    d.add_argument("name")
+    d.add_argument("--namespace", default="default")
+    d.add_argument("--server", default="http://127.0.0.1:8082")
+    d.set_defaults(fn=cmd_delete)
 
     args = p.parse_args()
     args.fn(args)

@@ -47,6 +47,14 @@ def add_admin_routes(app, store) -> None:
             return JSONResponse({"error": "not found"}, status_code=404)
         return obj
 
+    @app.delete("/admin/resources/{kind}/{name}")
+    def delete_resource(kind: str, name: str, namespace: str = "default"):
+        k = _canonical(kind)
+        if store.get(k, name, namespace) is None:
+            return JSONResponse({"error": f"{k} {name} not found"}, status_code=404)
+        store.delete(k, name, namespace)
+        return {"deleted": f"{k}/{name}"}
+
     @app.get("/admin/events/{name}")
     def get_events(name: str, namespace: str = "default"):
         return store.events_for(name, namespace)

@@ -94,3 +94,21 @@ def test_example_script_runs():
     )
     assert out.returncode == 0, out.stderr
     assert "FinalAnswer" in out.stdout
+
+
+def test_admin_delete_cascades(cp):
+    """DELETE /admin/resources/<kind>/<name> removes the object and its
+    owned children (the store's owner index)."""
+    client = TestClient(cp.rest_app)
+    for doc in yaml.safe_load_all(MANIFESTS):
+        client.post("/admin/resources", json=doc)
+    wait_for(
+        lambda: (cp.store.get(TASK, "yaml-task") or {}).get("status", {}).get("phase")
+        == TaskPhase.FINAL_ANSWER or None,
+        timeout=20,
+    )
+    r = client.delete("/admin/resources/Task/yaml-task")
+    assert r.status_code == 200 and r.json()["deleted"] == "Task/yaml-task"
+    assert cp.store.get(TASK, "yaml-task") is None
+    r = client.delete("/admin/resources/Task/yaml-task")
+    assert r.status_code == 404
