@@ -112,3 +112,13 @@ def test_admin_delete_cascades(cp):
     assert cp.store.get(TASK, "yaml-task") is None
     r = client.delete("/admin/resources/Task/yaml-task")
     assert r.status_code == 404
+
+
+def test_admin_rejects_malformed(cp):
+    client = TestClient(cp.rest_app)
+    r = client.post("/admin/resources", json={"kind": "Task"})  # no metadata.name
+    assert r.status_code == 400
+    r = client.get("/admin/resources/NoSuchKind")
+    assert r.status_code in (200, 404)  # empty list or not-found, never 500
+    r = client.delete("/admin/resources/Task/never-existed")
+    assert r.status_code == 404
