@@ -122,3 +122,50 @@ def test_admin_rejects_malformed(cp):
     assert r.status_code in (200, 404)  # empty list or not-found, never 500
     r = client.delete("/admin/resources/Task/never-existed")
     assert r.status_code == 404
+
+
+def test_serve_subprocess_boots_and_runs_loop(tmp_path):
+    """The actual `python -m agentcontrolplane_amd serve` CLI path: boot on
+    a free port with no engine, drive the mock-LLM loop over HTTP."""
+    import socket
+
+    import httpx
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "agentcontrolplane_amd", "serve",
+         "--port", str(port), "--device", "none", "--auto-approve",
+         "--wal", str(tmp_path / "wal.jsonl")],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+    )
+    try:
+        client = httpx.Client(base_url=f"http://127.0.0.1:{port}", timeout=5)
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            try:
+                if client.get("/status").status_code == 200:
+                    break
+            except Exception:
+                time.sleep(0.2)
+        else:
+            raise AssertionError("serve never came up")
+        for doc in yaml.safe_load_all(MANIFESTS):
+            r = client.post("/admin/resources", json=doc)
+            assert r.status_code == 201, r.text
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            t = client.get("/admin/resources/Task/yaml-task").json()
+            if t.get("status", {}).get("phase") == "FinalAnswer":
+                break
+            time.sleep(0.25)
+        else:
+            raise AssertionError(f"task never finished: {t.get('status', {})}")
+        assert t["status"]["output"] == "mock final answer"
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=5)
+        except subprocess.TimeoutExpired:
+            proc.kill()
