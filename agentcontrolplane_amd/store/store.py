@@ -106,15 +106,25 @@ class ResourceStore:
     # ------------------------------------------------------------------ WAL
 
     def _replay(self, path: str) -> None:
+        good_end = 0  # byte offset after the last valid line
+        with open(path, "rb") as f:
+            raw = f.read()
+        offset = 0
         with open(path, "r", encoding="utf-8") as f:
             for line in f:
+                raw_len = len(line.encode("utf-8"))
                 line = line.strip()
                 if not line:
+                    offset += raw_len
+                    good_end = offset
                     continue
                 try:
                     rec = json.loads(line)
                 except json.JSONDecodeError:
+                    offset += raw_len
                     continue  # torn tail write after a crash
+                offset += raw_len
+                good_end = offset
                 op, obj = rec.get("op"), rec.get("obj")
                 if not obj:
                     continue
@@ -129,6 +139,11 @@ class ResourceStore:
                 else:
                     bucket[name] = obj
                 self._rv = max(self._rv, int(m.get("resourceVersion", 0)))
+        # truncate away any torn tail so the next append starts a fresh
+        # line (appending onto a torn fragment would corrupt BOTH records)
+        if good_end < len(raw):
+            with open(path, "r+b") as f:
+                f.truncate(good_end)
         # rebuild the owner index from the replayed state
         self._owned_by.clear()
         for kind, nss in self._data.items():
