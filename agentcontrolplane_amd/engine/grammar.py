@@ -323,7 +323,9 @@ class SchemaArgsMachine:
         prog: List[tuple] = [("lit", b"{")]
         for i, (k, t) in enumerate(self.fields):
             sep = b", " if i else b""
-            prog.append(("lit", sep + b'"' + k.encode() + b'": '))
+            # json.dumps escapes quotes/backslashes/control bytes in the
+            # property name (a raw k.encode() would emit invalid JSON)
+            prog.append(("lit", sep + json.dumps(k).encode() + b": "))
             if t == "string":
                 prog.append(("str",))
             elif t in ("number", "integer"):
@@ -468,6 +470,10 @@ class ToolCallGrammar:
         if not tool_names:
             raise ValueError("no tools to constrain to")
         self.names = sorted(set(tool_names))
+        # trie operates on the JSON-escaped byte form (json.dumps minus the
+        # surrounding quotes): a name containing '"' or '\\' would
+        # otherwise emit invalid JSON
+        self._name_bytes = [json.dumps(n).encode()[1:-1] for n in self.names]
         self.buf = bytearray()
         self.phase = "pre"       # pre → name → mid → args → done
         self.pos = 0
@@ -490,8 +496,7 @@ class ToolCallGrammar:
             return {self.PRE[self.pos]}
         if self.phase == "name":
             nexts: Set[int] = set()
-            for n in self.names:
-                nb = n.encode()
+            for nb in self._name_bytes:
                 if nb.startswith(self.name_prefix):
                     if len(nb) > len(self.name_prefix):
                         nexts.add(nb[len(self.name_prefix)])
@@ -521,8 +526,10 @@ class ToolCallGrammar:
                 self.phase = "name"
             return
         if self.phase == "name":
-            if b == ord('"') and self.name_prefix.decode() in self.names:
-                self._select_args_machine(self.name_prefix.decode())
+            if b == ord('"') and self.name_prefix in self._name_bytes:
+                # unescape the trie form back to the raw tool name
+                raw = json.loads('"' + self.name_prefix.decode() + '"')
+                self._select_args_machine(raw)
                 self.phase = "mid"
                 self.pos = 1  # the '"' consumed is MID[0]... MID starts with '"'
                 return

@@ -130,3 +130,34 @@ def test_schema_free_form_fallback():
     g = ToolCallGrammar(tools=tools, max_args_len=60)
     obj = json.loads(drive(g, rng))
     assert isinstance(obj["arguments"], dict)
+
+
+def test_schema_keys_and_tool_names_with_json_specials():
+    """Property names and tool names containing quotes/backslashes must be
+    JSON-escaped in the forced output (raw emission produced invalid JSON)."""
+    import json
+    import random
+
+    from agentcontrolplane_amd.engine.tokenizer import EOT
+
+    cases = [
+        ({'a"b': {"type": "string"}}, "t__x"),
+        ({"a\\b": {"type": "integer"}}, "t__x"),
+        ({"ok": {"type": "boolean"}}, 'we"ird\\name'),
+    ]
+    for props, name in cases:
+        tools = [{"type": "function", "function": {
+            "name": name,
+            "parameters": {"type": "object", "properties": props,
+                           "required": list(props)}}}]
+        g = ToolCallGrammar(tools=tools, max_args_len=64)
+        rng = random.Random(3)
+        for _ in range(4000):
+            if g.accepting:
+                g.advance(EOT)
+                break
+            g.advance(rng.choice(sorted(g.allowed_tokens())))
+        n, args = g.parse()
+        assert n == name
+        parsed = json.loads(args)
+        assert set(parsed) == set(props)
