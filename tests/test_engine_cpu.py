@@ -268,3 +268,34 @@ def test_stream_close_cancels_request(engine):
     # engine still healthy
     r = engine.chat(msgs, sampling=SamplingParams(max_tokens=4, temperature=0))
     assert r.finish_reason in ("stop", "length")
+
+
+def test_no_kv_block_leak_after_workload():
+    """After all requests finish and the continuation cache is evicted,
+    every block must be back in the free pool."""
+    import threading
+
+    cfg = EngineConfig(
+        model="tiny", device="cpu", num_kv_blocks=256, kv_block_size=8,
+        max_prefill_tokens=64, request_timeout_s=120,
+    )
+    eng = InferenceEngine(cfg)
+    try:
+        def run(i):
+            eng.generate(
+                [(i * 7 + k) % 250 for k in range(10 + i * 3)],
+                SamplingParams(max_tokens=6 + i % 5, temperature=1.0, seed=i),
+            )
+        ts = [threading.Thread(target=run, args=(i,)) for i in range(20)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join()
+        assert not eng.scheduler.running and not eng.scheduler.waiting
+        # retired (continuation-cache) sequences hold the only references
+        while eng.scheduler.retired:
+            assert eng.scheduler._evict_one_retired()
+        assert eng.bm.used_blocks == 0, eng.bm.used_blocks
+        assert eng.bm.free_blocks == cfg.num_kv_blocks
+    finally:
+        eng.stop()
