@@ -575,3 +575,75 @@ def test_soak_concurrent_task_mix(cp):
     for t in done:
         assert t["status"]["output"] == "mock final answer"
         assert [m["role"] for m in t["status"]["contextWindow"]][-1] == "assistant"
+
+
+def test_workqueue_fuzz_no_concurrent_no_lost():
+    """_WorkQueue invariants under concurrent producers/consumers:
+    (a) a key is never handed to two workers at once,
+    (b) every key enqueued after its last processing is processed again
+        (the dirty-while-in-flight requeue)."""
+    import random
+    import threading
+    import time as _t
+
+    from agentcontrolplane_amd.controllers.manager import _WorkQueue
+
+    q = _WorkQueue()
+    keys = [("K", f"k{i}") for i in range(6)]
+    active = set()
+    active_lock = threading.Lock()
+    processed = {k: 0 for k in keys}
+    enqueued = {k: 0 for k in keys}
+    violations = []
+    stop = threading.Event()
+
+    def producer(seed):
+        rng = random.Random(seed)
+        for _ in range(300):
+            k = rng.choice(keys)
+            with active_lock:
+                enqueued[k] += 1
+            if rng.random() < 0.3:
+                q.add_after(k, rng.random() * 0.01)
+            else:
+                q.add(k)
+            if rng.random() < 0.1:
+                _t.sleep(0.001)
+
+    def worker():
+        while not stop.is_set():
+            k = q.get(timeout=0.05)
+            if k is None:
+                continue
+            with active_lock:
+                if k in active:
+                    violations.append(k)
+                active.add(k)
+            _t.sleep(0.0005)
+            with active_lock:
+                active.discard(k)
+                processed[k] += 1
+            q.done(k)
+
+    workers = [threading.Thread(target=worker) for _ in range(4)]
+    producers = [threading.Thread(target=producer, args=(i,)) for i in range(3)]
+    for t in workers + producers:
+        t.start()
+    for t in producers:
+        t.join()
+    # drain: wait until nothing is pending
+    deadline = _t.monotonic() + 10
+    while _t.monotonic() < deadline:
+        with q._cv:
+            idle = not q._ready and not q._delayed and not q._in_flight
+        if idle:
+            break
+        _t.sleep(0.02)
+    stop.set()
+    for t in workers:
+        t.join()
+    assert not violations, f"concurrent reconcile of {violations[:3]}"
+    for k in keys:
+        assert processed[k] >= 1, (k, enqueued[k])
+    with q._cv:
+        assert not q._dirty_while_in_flight
