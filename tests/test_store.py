@@ -218,3 +218,36 @@ def test_store_replay_equivalence_random_ops():
             for k in want:
                 assert want[k]["spec"] == got[k]["spec"], (trial, k)
                 assert want[k]["metadata"]["resourceVersion"] == got[k]["metadata"]["resourceVersion"]
+
+
+def test_lease_contention_fuzz(store):
+    """Two 'pods' hammer the same lease with short durations: at no instant
+    do both believe they hold it (mutual exclusion incl. expiry takeover)."""
+    import threading
+    import time as _t
+
+    holders = set()
+    lock = threading.Lock()
+    violations = []
+    stop = _t.monotonic() + 3.0
+
+    def pod(name):
+        while _t.monotonic() < stop:
+            if store.acquire_lease("race", name, duration_s=0.05):
+                with lock:
+                    if holders:
+                        violations.append((name, set(holders)))
+                    holders.add(name)
+                _t.sleep(0.01)  # hold briefly (well under the duration)
+                with lock:
+                    holders.discard(name)
+                store.release_lease("race", name)
+            else:
+                _t.sleep(0.002)
+
+    ts = [threading.Thread(target=pod, args=(n,)) for n in ("pod-a", "pod-b")]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not violations, violations[:3]
