@@ -221,8 +221,11 @@ def test_store_replay_equivalence_random_ops():
 
 
 def test_lease_contention_fuzz(store):
-    """Two 'pods' hammer the same lease with short durations: at no instant
-    do both believe they hold it (mutual exclusion incl. expiry takeover)."""
+    """Two 'pods' hammer the same lease: at no instant do both believe they
+    hold it.  The duration (2 s) dwarfs the hold time, as in the reference
+    (LEASE_DURATION 30 s vs sub-second critical sections) — with a duration
+    comparable to scheduler jitter, expiry-based takeover of a STALLED
+    holder is legal lease semantics, not a violation."""
     import threading
     import time as _t
 
@@ -233,7 +236,7 @@ def test_lease_contention_fuzz(store):
 
     def pod(name):
         while _t.monotonic() < stop:
-            if store.acquire_lease("race", name, duration_s=0.05):
+            if store.acquire_lease("race", name, duration_s=2.0):
                 with lock:
                     if holders:
                         violations.append((name, set(holders)))
