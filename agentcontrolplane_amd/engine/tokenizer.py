@@ -51,6 +51,11 @@ class ByteTokenizer:
     ) -> List[int]:
         ids = [BOS]
         tools = tools or []
+        if tools and not (messages and messages[0].get("role") == "system"):
+            # no system turn to carry the tool schemas: synthesize one, else
+            # the model never sees the tools (OpenAI-endpoint calls often
+            # have no system message)
+            messages = [{"role": "system", "content": "You can call tools."}] + list(messages)
         for i, m in enumerate(messages):
             role = m.get("role", "user")
             content = m.get("content", "")
