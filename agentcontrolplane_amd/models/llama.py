@@ -80,19 +80,26 @@ class LlamaForCausalLM:
     def random_init(self, seed: int = 0) -> None:
         """Random-init weights of the real shapes (synthetic bench: no
         network for checkpoints — BASELINE.md).  Init in manageable chunks
-        directly on the device."""
-        g = torch.Generator(device=self.device).manual_seed(seed + self.tp_rank)
+        directly on the device.
+
+        TP consistency: REPLICATED tensors (embed, lm_head, norms) must be
+        bit-identical on every rank — they feed the all-reduced residual
+        stream — so they come from a rank-independent generator.  SHARDED
+        tensors are each rank's own slice of the logical full model, so a
+        rank-offset seed just picks which random slice this rank holds."""
+        g_rep = torch.Generator(device=self.device).manual_seed(seed)
+        g = torch.Generator(device=self.device).manual_seed(seed + 1 + self.tp_rank)
         cfg = self.cfg
         h = cfg.hidden_size
         std = 0.02
 
-        def randw(*shape):
+        def randw(*shape, gen=None):
             w = torch.empty(shape, dtype=self.dtype, device=self.device)
-            w.normal_(0.0, std, generator=g)
+            w.normal_(0.0, std, generator=gen or g)
             return w
 
-        self.embed = randw(cfg.vocab_size, h)
-        self.lm_head = self.embed if cfg.tie_embeddings else randw(cfg.vocab_size, h)
+        self.embed = randw(cfg.vocab_size, h, gen=g_rep)
+        self.lm_head = self.embed if cfg.tie_embeddings else randw(cfg.vocab_size, h, gen=g_rep)
         self.final_norm = torch.ones(h, dtype=self.dtype, device=self.device)
         qd = self.n_heads * self.head_dim
         kvd = self.n_kv_heads * self.head_dim

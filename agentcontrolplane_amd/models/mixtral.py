@@ -35,13 +35,16 @@ class MixtralForCausalLM(LlamaForCausalLM):
 
     def random_init(self, seed: int = 0) -> None:
         super().random_init(seed)
+        # routers are REPLICATED (every rank must route identically in EP);
+        # expert weights are sharded/owned per rank
+        g_rep = torch.Generator(device=self.device).manual_seed(seed + 500)
         g = torch.Generator(device=self.device).manual_seed(seed + 1000 + self.tp_rank)
         h = self.cfg.hidden_size
         std = 0.02
 
-        def randw(*shape):
+        def randw(*shape, gen=None):
             w = torch.empty(shape, dtype=self.dtype, device=self.device)
-            w.normal_(0.0, std, generator=g)
+            w.normal_(0.0, std, generator=gen or g)
             return w
 
         self.routers = []
@@ -51,7 +54,7 @@ class MixtralForCausalLM(LlamaForCausalLM):
             # the dense MLP weights are replaced per-expert; free them
             lw.gate_up = None
             lw.down = None
-            self.routers.append(randw(self.num_experts, h))
+            self.routers.append(randw(self.num_experts, h, gen=g_rep))
             self.expert_gate_up.append(randw(self.num_experts, 2 * self.inter, h))
             self.expert_down.append(randw(self.num_experts, h, self.inter))
 

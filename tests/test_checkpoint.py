@@ -66,3 +66,27 @@ def test_mixtral_checkpoint_roundtrip(tmp_path):
     l1 = m1.forward(_batch(8, cfg.vocab_size))
     l2 = m2.forward(_batch(8, cfg.vocab_size))
     assert torch.equal(l1, l2)
+
+
+def test_random_init_tp_replicated_consistency():
+    """Replicated tensors (embed, lm_head, routers) must be identical
+    across TP ranks; sharded tensors are each rank's own slice."""
+    from agentcontrolplane_amd.engine.config import PRESETS, EngineConfig
+    from agentcontrolplane_amd.models.llama import LlamaForCausalLM
+    from agentcontrolplane_amd.models.mixtral import MixtralForCausalLM
+
+    ecfg = EngineConfig(model="tiny", device="cpu")
+    r0 = LlamaForCausalLM(PRESETS["tiny"], ecfg, "cpu", tp_rank=0, tp_world=2)
+    r1 = LlamaForCausalLM(PRESETS["tiny"], ecfg, "cpu", tp_rank=1, tp_world=2)
+    r0.random_init(7)
+    r1.random_init(7)
+    assert torch.equal(r0.embed, r1.embed)
+    assert torch.equal(r0.lm_head, r1.lm_head)
+    assert not torch.equal(r0.layers[0].qkv, r1.layers[0].qkv)
+
+    m0 = MixtralForCausalLM(PRESETS["tiny-moe"], ecfg, "cpu", tp_rank=0, tp_world=2)
+    m1 = MixtralForCausalLM(PRESETS["tiny-moe"], ecfg, "cpu", tp_rank=1, tp_world=2)
+    m0.random_init(7)
+    m1.random_init(7)
+    assert torch.equal(m0.routers[0], m1.routers[0])
+    assert not torch.equal(m0.expert_gate_up[0], m1.expert_gate_up[0])
