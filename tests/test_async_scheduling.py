@@ -191,3 +191,67 @@ def test_seeded_and_penalized_equivalence():
             eng.stop()
 
     assert run(False) == run(True)
+
+
+def test_long_soak_staggered_mixed_arrivals():
+    """Staggered (non-wave) arrivals of free/tool/seeded/abandoned-stream
+    requests under a tiny pool: no errors, no KV leak, and the pipeline
+    keeps engaging (spec_steps > 0) despite the stateful-sampling mix."""
+    import json
+    import random
+    import time as _t
+
+    eng = make_engine(True, num_kv_blocks=96, kv_block_size=8,
+                      max_prefill_tokens=96)
+    params = {"type": "object", "properties": {"m": {"type": "string"}},
+              "required": ["m"]}
+    tools = [{"type": "function", "function": {"name": "s__t", "parameters": params}}]
+    errors = []
+    lock = threading.Lock()
+    rng = random.Random(424242)
+
+    def job(i, kind):
+        try:
+            if kind == "tool":
+                r = eng.chat(
+                    [{"role": "user", "content": "x" * rng.randrange(5, 60)}],
+                    tools=tools,
+                    sampling=SamplingParams(max_tokens=48, temperature=0.8,
+                                            tool_choice="required"),
+                )
+                json.loads(r.tool_calls[0]["function"]["arguments"])
+            elif kind == "stream":
+                for n, _ in enumerate(eng.chat_stream(
+                    [{"role": "user", "content": "s" * 20}],
+                    sampling=SamplingParams(max_tokens=24, temperature=1.0),
+                )):
+                    if n > 4:
+                        break  # abandon → cancellation path
+            else:
+                eng.generate(
+                    [rng.randrange(250) for _ in range(rng.randrange(5, 80))],
+                    SamplingParams(max_tokens=rng.randrange(2, 20),
+                                   temperature=1.0, seed=i,
+                                   frequency_penalty=0.3),
+                )
+        except Exception as e:  # noqa: BLE001
+            with lock:
+                errors.append((i, kind, repr(e)))
+
+    threads = []
+    for i in range(60):
+        kind = rng.choice(["tool", "stream", "free", "free"])
+        t = threading.Thread(target=job, args=(i, kind))
+        t.start()
+        threads.append(t)
+        _t.sleep(rng.random() * 0.01)
+    for t in threads:
+        t.join()
+    assert not errors, errors[:4]
+    _t.sleep(0.5)  # let stream cancellations drain
+    while eng.scheduler.retired:
+        assert eng.scheduler._evict_one_retired()
+    assert not eng.scheduler.running and not eng.scheduler.waiting
+    assert eng.bm.used_blocks == 0
+    assert eng.metrics().get("spec_steps", 0) > 0
+    eng.stop()
