@@ -78,3 +78,27 @@ def test_inproc_registry(store):
     tools = mgr.connect_server(obj)
     assert tools[0]["name"] == "double"
     assert mgr.call_tool("local", "double", {"x": 4}) == "8.0"
+
+
+def test_stdio_concurrent_calls(store):
+    """Many threads calling one stdio server: the per-connection lock
+    serializes frames — every call gets ITS OWN response."""
+    import threading
+
+    mgr = MCPServerManager(store)
+    try:
+        mgr.connect_server(_stdio_server_obj())
+        results = [None] * 24
+
+        def go(i):
+            results[i] = mgr.call_tool("calc", "add", {"a": i, "b": 1000})
+
+        ts = [threading.Thread(target=go, args=(i,)) for i in range(24)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join()
+        for i, r in enumerate(results):
+            assert r == str(float(i + 1000)), (i, r)
+    finally:
+        mgr.close()
