@@ -26,6 +26,7 @@ Differences by design (MI355X-first):
 """
 from __future__ import annotations
 
+import json
 import threading
 import time
 from typing import Dict, List, Optional, Tuple
@@ -655,7 +656,7 @@ class TaskReconciler(Reconciler):
                         "id": tc_id,
                         "function": {
                             "name": "respond_to_human",
-                            "arguments": __import__("json").dumps({"content": output.content}),
+                            "arguments": json.dumps({"content": output.content}),
                         },
                         "type": "function",
                     }
@@ -688,7 +689,7 @@ class TaskReconciler(Reconciler):
                     "taskRef": {"name": name},
                     "toolRef": {"name": "respond_to_human"},
                     "toolType": ToolType.HUMAN_CONTACT,
-                    "arguments": __import__("json").dumps({"content": output.content}),
+                    "arguments": json.dumps({"content": output.content}),
                 },
                 "status": {},
             }
