@@ -108,3 +108,41 @@ def test_rmsnorm_and_swiglu_oracles_vs_brute_force():
     u = gu[:, 32:].numpy().astype(np.float64)
     want = (g / (1 + np.exp(-g))) * u
     assert np.abs(got - want).max() < 1e-4
+
+
+def test_rope_oracle_vs_brute_force():
+    """Rotate-half RoPE: pair (x[i], x[i+D/2]) rotates by pos·θ^(-2i/D)."""
+    N, H, D, P = 6, 2, 16, 40
+    theta = 10000.0
+    torch.manual_seed(4)
+    q = torch.randn(N, H, D)
+    k = torch.randn(N, 1, D)
+    v = torch.randn(N, 1, D)
+    q0, k0 = q.numpy().astype(np.float64).copy(), k.numpy().astype(np.float64).copy()
+    pos = torch.tensor([0, 1, 5, 17, 2, 39], dtype=torch.long)
+    nb = 8
+    kc = torch.zeros(nb, 4, 1, D)
+    vc = torch.zeros(nb, 4, 1, D)
+    slots = torch.tensor([3, 9, 12, 20, 25, 31], dtype=torch.long)
+    cs = ref.build_cos_sin(P, D, theta, "cpu")
+    ref.rope_and_cache(q, k, v, pos, slots, kc, vc, cs)
+    for n in range(N):
+        p = int(pos[n])
+        for h in range(H):
+            for i in range(D // 2):
+                ang = p / theta ** (2 * i / D)
+                c, s = math.cos(ang), math.sin(ang)
+                want_lo = q0[n, h, i] * c - q0[n, h, i + D // 2] * s
+                want_hi = q0[n, h, i + D // 2] * c + q0[n, h, i] * s
+                assert abs(float(q[n, h, i]) - want_lo) < 1e-4, (n, h, i)
+                assert abs(float(q[n, h, i + D // 2]) - want_hi) < 1e-4
+    # K landed rope'd at the right flat slots; V raw
+    flat_k = kc.view(-1, 1, D)
+    flat_v = vc.view(-1, 1, D)
+    for n in range(N):
+        p = int(pos[n])
+        ang = p / theta ** 0.0  # i = 0 pair
+        c, s = math.cos(ang), math.sin(ang)
+        want = k0[n, 0, 0] * c - k0[n, 0, D // 2] * s
+        assert abs(float(flat_k[slots[n], 0, 0]) - want) < 1e-4
+        assert torch.allclose(flat_v[slots[n]], v[n])
