@@ -128,7 +128,10 @@ def main() -> None:
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--model", default="llama3-8b")
-    p.add_argument("--concurrency", type=int, default=128, help="tasks per GPU")
+    p.add_argument(
+        "--concurrency", type=int, default=1000,
+        help="tasks per GPU (default = the 1k-concurrent-Tasks config BASELINE.json names)",
+    )
     p.add_argument("--decode-tokens", type=int, default=32)
     p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree (else DP sharding)")
     p.add_argument("--device", default=None)
