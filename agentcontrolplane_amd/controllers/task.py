@@ -41,7 +41,6 @@ from ..api.types import (
     Message,
     TaskPhase,
     TaskStatusType,
-    ToolCallPhase,
     ToolType,
     owner_ref,
 )

@@ -13,7 +13,7 @@ writes land nowhere meaningful, and their logits are never sampled.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 import torch
 

@@ -13,7 +13,7 @@ as the numerics oracle for the C++ one (tests assert identical behavior).
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 
 class OutOfBlocksError(RuntimeError):

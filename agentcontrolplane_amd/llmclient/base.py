@@ -11,7 +11,7 @@ controller tests GPU-free, mirroring the reference's mockgen seam
 from __future__ import annotations
 
 import dataclasses
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 from ..api.types import Message, ToolType
 

@@ -26,7 +26,7 @@ from typing import Optional
 import torch
 import torch.distributed as dist
 
-from ..engine.batch import batch_from_wire, batch_to_wire
+from ..engine.batch import batch_from_wire
 
 STOP = {"op": "stop"}
 

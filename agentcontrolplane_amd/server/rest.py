@@ -27,7 +27,7 @@ the HumanLayer SaaS, and /metrics exposes engine/controller counters.
 from __future__ import annotations
 
 import json
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 from fastapi import FastAPI, Request, Response
 from fastapi.responses import JSONResponse

@@ -29,7 +29,7 @@ import os
 import queue
 import threading
 import time
-from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple
+from typing import Any, Dict, Iterable, List, Optional, Tuple
 
 from ..api.types import API_VERSION, EVENT, LEASE, new_object_meta, now_iso
 

@@ -29,7 +29,6 @@ import random
 import statistics
 import string
 import sys
-import threading
 import time
 
 
