@@ -26,7 +26,7 @@ import time
 import traceback
 from typing import Dict, List, Optional, Tuple
 
-from ..store import ResourceStore, WatchEvent
+from ..store import ConflictError, NotFoundError, ResourceStore, WatchEvent
 
 logger = logging.getLogger("acp.manager")
 
@@ -209,8 +209,13 @@ class ControllerManager:
         if not name:
             return
         for rec in self._reconcilers:
-            if rec.kind == ev.kind and ev.type != "DELETED":
-                self._queues[rec.kind].add((name, ns))
+            if rec.kind == ev.kind:
+                if ev.type != "DELETED":
+                    self._queues[rec.kind].add((name, ns))
+                else:
+                    hook = getattr(rec, "on_deleted", None)
+                    if hook is not None:
+                        hook(name, ns)
             if ev.kind in rec.owns:
                 mapped = rec.map_owned(ev)
                 if mapped is not None:
@@ -229,6 +234,19 @@ class ControllerManager:
                 q.done(key)
                 if res is not None and (res.requeue or res.requeue_after > 0):
                     q.add_after(key, res.requeue_after)
+            except NotFoundError:
+                # expected race: the object was deleted while a reconcile
+                # was in flight — the reference's client.IgnoreNotFound
+                # (e.g. task_controller.go Reconcile's Get error path);
+                # drop the key, no traceback, no requeue
+                self._error_backoff.pop((rec.kind, name, ns), None)
+                q.done(key)
+            except ConflictError:
+                # optimistic-concurrency conflict: re-read and retry, the
+                # apiserver-409 path the reference handles with
+                # RetryOnConflict loops — requeue quickly, quietly
+                q.done(key)
+                q.add_after(key, 0.01)
             except Exception:
                 logger.error(
                     "reconcile %s %s/%s failed:\n%s", rec.kind, ns, name, traceback.format_exc()

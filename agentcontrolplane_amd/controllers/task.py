@@ -93,6 +93,15 @@ class TaskReconciler(Reconciler):
         self._inflight: Dict[Tuple[str, str], dict] = {}
         self._manager = None  # back-ref set by ControllerManager.register
 
+    def on_deleted(self, name: str, namespace: str) -> None:
+        """Drop per-task state when the Task object goes away — the mutex
+        and in-flight maps otherwise grow one entry per task forever under
+        create/delete churn."""
+        key = (namespace, name)
+        self._inflight.pop(key, None)
+        with self._mutex_guard:
+            self._mutexes.pop(key, None)
+
     # ToolCall events requeue the parent task via the task label
     def map_owned(self, ev):
         labels = ev.obj.get("metadata", {}).get("labels", {}) or {}

@@ -230,7 +230,28 @@ class ToolCallReconciler(Reconciler):
             return self._execute_delegate(tc)
         if tool_type == ToolType.HUMAN_CONTACT:
             return self._execute_human_contact(tc)
-        return self._execute_mcp(tc)
+        return self._execute_external(tc) or self._execute_mcp(tc)
+
+    def _execute_external(self, tc) -> Optional[Result]:
+        """Optional external-API path (externalAPI/main.go:32-67 is unused
+        scaffolding in the reference; here a ToolCall whose toolRef matches a
+        registered client executes through the registry instead of MCP).
+        Returns None when the tool is not registered."""
+        from ..external_api import DEFAULT_REGISTRY
+
+        tool_name = tc["spec"].get("toolRef", {}).get("name", "")
+        if not DEFAULT_REGISTRY.has(tool_name):
+            return None
+        try:
+            args = json.loads(tc["spec"].get("arguments", "") or "{}")
+        except json.JSONDecodeError as e:
+            return self._fail(tc, f"invalid arguments JSON: {e}")
+        try:
+            client = DEFAULT_REGISTRY.get_client(tool_name)
+            result = client.call(args)
+        except Exception as e:
+            return self._fail(tc, str(e))
+        return self._finish(tc, result)
 
     def _finish(self, tc, result: str) -> Result:
         status = tc["status"]

@@ -86,7 +86,9 @@ class InferenceEngine:
                 min(config.max_model_len, self.mcfg.max_position)
                 + config.kv_block_size - 1
             ) // config.kv_block_size
-            max_graph_batch = min(config.max_batch_size, 512)
+            # 1024 covers the decode-only tail of the 1k-concurrent-Task
+            # bench (512 left ~half those steps on the eager path)
+            max_graph_batch = min(config.max_batch_size, 1024)
             if moe_graph_cap is not None:
                 max_graph_batch = min(max_graph_batch, moe_graph_cap)
             self.graph_runner = DecodeGraphRunner(
@@ -96,10 +98,21 @@ class InferenceEngine:
                 scratch_block=num_blocks,
                 kv_block_size=config.kv_block_size,
             )
+        self._ctx_limit = min(config.max_model_len, self.mcfg.max_position)
         self._gen = torch.Generator(device=self.device.type)
         self._gen.manual_seed(config.seed)
         self._sampling_cache_key = None
         self._sampling_cache = None
+        # persistent pinned→device staging for per-step sampling metadata
+        # (see stage.PersistentStage: unpinned rebuilds were ~10 ms/step)
+        import numpy as _np
+
+        cap = config.max_batch_size + 8
+        stager = self.scheduler.stager
+        self._ps_temp = stager.persistent(cap, (), _np.float32)
+        self._ps_topk = stager.persistent(cap, (), _np.int64)
+        self._ps_topp = stager.persistent(cap, (), _np.float32)
+        self._ps_mask = stager.persistent(cap, (N_SPECIAL,), _np.bool_)
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
         self._pending: List[InferenceRequest] = []
@@ -193,11 +206,14 @@ class InferenceEngine:
     # ------------------------------------------------------------- submit
 
     def submit(self, request: InferenceRequest) -> InferenceRequest:
+        from .scheduler import worst_case_output_tokens
+
         limit = min(self.cfg.max_model_len, self.mcfg.max_position)
-        if len(request.prompt_ids) + request.sampling.max_tokens > limit:
+        worst = worst_case_output_tokens(request)
+        if len(request.prompt_ids) + worst > limit:
             raise ValueError(
-                f"request length {len(request.prompt_ids)}+{request.sampling.max_tokens} "
-                f"exceeds the model context limit {limit}"
+                f"request length {len(request.prompt_ids)}+{worst} (worst-case "
+                f"emission) exceeds the model context limit {limit}"
             )
         with self._work:
             if len(self._pending) >= self.cfg.max_queue:
@@ -241,7 +257,16 @@ class InferenceEngine:
                 constrained = not has_tool_result
             elif choice == "required":
                 constrained = True
-        return InferenceRequest(prompt_ids, sampling, constrained, tools)
+        pre_in_prompt = False
+        if constrained:
+            # the tool-call JSON always opens with the forced PRE bytes:
+            # emitting them as prompt tokens turns len(PRE) sequential
+            # masked decode steps into one chunked-prefill extension
+            from .grammar import ToolCallGrammar
+
+            prompt_ids = prompt_ids + list(ToolCallGrammar.PRE)
+            pre_in_prompt = True
+        return InferenceRequest(prompt_ids, sampling, constrained, tools, pre_in_prompt)
 
     def _result_of(self, req: InferenceRequest, t0: float) -> ChatResult:
         out_ids = req.output_ids
@@ -497,40 +522,39 @@ class InferenceEngine:
         live = logits[:, :N_SPECIAL]  # sampling restricted to decodable ids
         self._m["ls_gather_s"] = self._m.get("ls_gather_s", 0.0) + time.monotonic() - ta
         tb = time.monotonic()
-        # per-sequence sampling params are static: cache the device tensors
-        # against the batch's id tuple (stable across decode steps)
+        # per-sequence sampling params are static: refill the persistent
+        # staging buffers only when the batch's id tuple changes (a row's
+        # sequence may have finished at the commit that just ran — sample
+        # it with defaults, the commit discards its token)
         key = tuple(batch.sample_seq_ids)
-        cached = self._sampling_cache if self._sampling_cache_key == key else None
-        if cached is None:
-            # fresh device tensors staged through pinned memory: a plain
-            # torch.tensor(..., device="cuda") is a BLOCKING copy that
-            # drains the speculative pipeline (profiles/r01: 5.2 s/bench)
-            # a row's sequence may have finished at the commit that just
-            # ran (deferred grammar speculation): sample it with defaults,
-            # the commit discards its token
-            stager = self.scheduler.stager
-            sp = [s.request.sampling if s is not None else None for s in seqs]
-            temps = stager.fresh([p.temperature if p else 0.0 for p in sp], "float32")
-            top_ks = stager.fresh([p.top_k if p else 0 for p in sp], "int64")
-            top_ps = stager.fresh([p.top_p if p else 1.0 for p in sp], "float32")
+        if self._sampling_cache_key != key:
+            ht, hk, hp = self._ps_temp.host(), self._ps_topk.host(), self._ps_topp.host()
+            for i, s in enumerate(seqs):
+                p = s.request.sampling if s is not None else None
+                ht[i] = p.temperature if p else 0.0
+                hk[i] = p.top_k if p else 0
+                hp[i] = p.top_p if p else 1.0
             self._sampling_cache_key = key
-            self._sampling_cache = (temps, top_ks, top_ps)
-        else:
-            temps, top_ks, top_ps = cached
+            self._sampling_cache = (
+                self._ps_temp.commit(B),
+                self._ps_topk.commit(B),
+                self._ps_topp.commit(B),
+            )
+        temps, top_ks, top_ps = self._sampling_cache
         mask = None
         if any(s is not None and s.grammar is not None for s in seqs):
-            import numpy as np
-
             tm = time.monotonic()
-            m = np.ones((B, N_SPECIAL), dtype=bool)
+            m = self._ps_mask.host()
+            m[:B] = True
             for i, s in enumerate(seqs):
                 if s is not None and s.grammar is not None:
                     allowed = s.grammar.allowed_tokens()
                     if s.grammar.accepting:
                         allowed = set(allowed) | {EOT}
-                    m[i] = False
-                    m[i, list(allowed)] = True
-            mask = self.scheduler.stager.tensor("mask", m, "bool")
+                    row = m[i]
+                    row[:] = False
+                    row[list(allowed)] = True
+            mask = self._ps_mask.commit(B)
             self._m["mask_time_s"] = (
                 self._m.get("mask_time_s", 0.0) + time.monotonic() - tm
             )
@@ -603,8 +627,11 @@ class InferenceEngine:
                     self.scheduler.finish_seq(s, "tool_calls")
                     continue
                 s.grammar.advance(tok)
-                if len(s.request.output_ids) >= s.request.sampling.max_tokens * 8:
-                    # runaway guard; grammar's closing mode should prevent this
+                if len(s.request.output_ids) >= s.request.sampling.max_tokens * 8 or (
+                    s.total_len >= self._ctx_limit
+                ):
+                    # runaway guard + hard context-limit stop: positions must
+                    # never walk past the RoPE table (cfg.max_position)
                     self._m["requests_completed"] += 1
                     self.scheduler.finish_seq(s, "length")
                 continue
@@ -616,7 +643,9 @@ class InferenceEngine:
                 continue
             if s.request.on_token is not None:
                 s.request.on_token(tok)
-            if len(s.request.output_ids) >= s.request.sampling.max_tokens:
+            if len(s.request.output_ids) >= s.request.sampling.max_tokens or (
+                s.total_len >= self._ctx_limit
+            ):
                 # request-level count: survives recompute preemption (the
                 # per-sequence list folds into the prompt on preempt)
                 self._m["requests_completed"] += 1

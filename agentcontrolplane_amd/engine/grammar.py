@@ -457,7 +457,7 @@ class ToolCallGrammar:
     MID = b'", "arguments": '
 
     def __init__(self, tool_names: Optional[List[str]] = None, max_args_len: int = 2048,
-                 tools: Optional[List[dict]] = None):
+                 tools: Optional[List[dict]] = None, pre_in_prompt: bool = False):
         self.schemas: Dict[str, dict] = {}
         if tools:
             names = []
@@ -474,8 +474,14 @@ class ToolCallGrammar:
         # surrounding quotes): a name containing '"' or '\\' would
         # otherwise emit invalid JSON
         self._name_bytes = [json.dumps(n).encode()[1:-1] for n in self.names]
-        self.buf = bytearray()
-        self.phase = "pre"       # pre → name → mid → args → done
+        if pre_in_prompt:
+            # the engine emitted PRE as prompt tokens (one prefill chunk):
+            # parsing must still see the full JSON, so pre-seed the buffer
+            self.buf = bytearray(self.PRE)
+            self.phase = "name"
+        else:
+            self.buf = bytearray()
+            self.phase = "pre"   # pre → name → mid → args → done
         self.pos = 0
         self.max_args_len = max_args_len
         self.name_prefix = b""

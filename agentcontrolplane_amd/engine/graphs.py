@@ -19,7 +19,7 @@ import torch
 
 from .batch import FlatBatch
 
-BUCKETS = [4, 8, 16, 32, 64, 96, 128, 192, 256, 384, 512]
+BUCKETS = [4, 8, 16, 32, 64, 96, 128, 192, 256, 384, 512, 768, 1024]
 
 
 class DecodeGraphRunner:

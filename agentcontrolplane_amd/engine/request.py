@@ -40,13 +40,18 @@ class InferenceRequest:
     _counter_lock = threading.Lock()
 
     def __init__(self, prompt_ids: List[int], sampling: SamplingParams,
-                 constrained: bool = False, tools: Optional[List[Dict]] = None):
+                 constrained: bool = False, tools: Optional[List[Dict]] = None,
+                 pre_in_prompt: bool = False):
         with self._counter_lock:
             self._counter[0] += 1
             self.request_id = self._counter[0]
         self.prompt_ids = prompt_ids
         self.sampling = sampling
         self.constrained = constrained
+        # True when the caller appended ToolCallGrammar.PRE ('{"name": "')
+        # to the prompt: the forced preamble then costs one prefill chunk
+        # instead of len(PRE) sequential masked decode steps
+        self.pre_in_prompt = pre_in_prompt
         self.tools = tools or []
         self.output_ids: List[int] = []
         self.submit_time = time.monotonic()

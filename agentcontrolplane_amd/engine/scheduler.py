@@ -16,6 +16,7 @@ cache).
 from __future__ import annotations
 
 import dataclasses
+import json
 from typing import Dict, List, Optional, Tuple
 
 import torch
@@ -26,6 +27,27 @@ from .grammar import ToolCallGrammar
 from .request import InferenceRequest
 
 WAITING, PREFILL, DECODE, FINISHED = "waiting", "prefill", "decode", "finished"
+
+
+def worst_case_output_tokens(request: InferenceRequest) -> int:
+    """Upper bound on tokens a request can emit.
+
+    Free-running requests stop at ``max_tokens`` exactly.  Constrained
+    requests can overshoot it: the grammar must finish the scaffolding
+    (name + MID) and legally unwind the arguments object, so admission /
+    KV reservation / context-limit checks must budget for that instead of
+    trusting ``max_tokens`` (overshoot near the context limit
+    walked positions past the RoPE table)."""
+    sp = request.sampling
+    if not request.constrained or not request.tools:
+        return sp.max_tokens
+    names = [t["function"]["name"] for t in request.tools]
+    max_name = max(len(json.dumps(n)) - 2 for n in names)
+    max_args = max(16, sp.max_tokens - 24 - max(len(n) for n in names))
+    pre = 0 if getattr(request, "pre_in_prompt", False) else len(ToolCallGrammar.PRE)
+    # pre + name + closing quote + MID + args + unwind slack (close braces,
+    # literal completion) + EOT
+    return pre + max_name + 1 + len(ToolCallGrammar.MID) + max_args + 16
 
 
 class Sequence:
@@ -49,7 +71,11 @@ class Sequence:
             # the grammar's closing mode needs ~20 tokens of slack for the
             # name/scaffolding plus the shortest legal unwind
             max_args = max(16, request.sampling.max_tokens - 24 - max(len(n) for n in names))
-            self.grammar = ToolCallGrammar(tools=request.tools, max_args_len=max_args)
+            self.grammar = ToolCallGrammar(
+                tools=request.tools,
+                max_args_len=max_args,
+                pre_in_prompt=request.pre_in_prompt,
+            )
 
     @property
     def total_len(self) -> int:
@@ -196,7 +222,9 @@ class Scheduler:
         # then admit waiting sequences
         while self.waiting and budget > 0 and len(self.running) < self.cfg.max_batch_size:
             s = self.waiting[0]
-            total_blocks = (s.total_len + s.request.sampling.max_tokens + bs - 1) // bs
+            total_blocks = (
+                s.total_len + worst_case_output_tokens(s.request) + bs - 1
+            ) // bs
             if total_blocks > self.bm.num_blocks:
                 self.waiting.pop(0)
                 self.abort(

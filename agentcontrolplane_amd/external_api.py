@@ -41,3 +41,5 @@ class Registry:
 
 
 default_registry = Registry()
+# reference naming (externalAPI/main.go:67 DefaultRegistry)
+DEFAULT_REGISTRY = default_registry

@@ -58,6 +58,12 @@ def _llama_name_map(model) -> Dict[str, torch.Tensor]:
 def save_checkpoint(model, path: str, max_shard_bytes: int = 4 << 30) -> None:
     from safetensors.torch import save_file
 
+    if getattr(model, "tp_world", 1) > 1:
+        # each rank holds only its shard under the full HF tensor name;
+        # writing those would silently produce a corrupt checkpoint
+        raise ValueError(
+            "save_checkpoint requires tp_world == 1 (gather shards first)"
+        )
     os.makedirs(path, exist_ok=True)
     tensors = _llama_name_map(model)
     shards, cur, cur_bytes = [], {}, 0

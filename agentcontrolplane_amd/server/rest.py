@@ -424,7 +424,9 @@ def build_app(store, manager=None, engine=None) -> FastAPI:
                 {"error": {"message": str(e), "type": "server_error"}}, status_code=500
             )
         msg: Dict[str, Any] = {"role": "assistant", "content": result.text or None}
-        finish = "stop" if result.finish_reason in ("stop", "length") else result.finish_reason
+        # 'length' passes through unchanged (SSE path already preserves it;
+        # masking it as 'stop' hid truncation from OpenAI clients)
+        finish = result.finish_reason
         if result.tool_calls:
             msg["tool_calls"] = result.tool_calls
             msg["content"] = None

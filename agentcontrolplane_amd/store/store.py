@@ -22,7 +22,6 @@ Status is a subresource: ``update_status`` only replaces ``status`` (like
 """
 from __future__ import annotations
 
-import copy
 import dataclasses
 import json
 import os
@@ -32,6 +31,20 @@ import time
 from typing import Any, Dict, Iterable, List, Optional, Tuple
 
 from ..api.types import API_VERSION, EVENT, LEASE, new_object_meta, now_iso
+
+
+def _copy(o):
+    """Fast deep copy for JSON-shaped data (dict/list/scalars only).
+
+    ``copy.deepcopy`` pays memo bookkeeping on every node; store objects are
+    pure JSON trees, so a recursive rebuild is ~4x faster — this runs on
+    every get/list/update under the store lock, so it is hot at 1k
+    concurrent Tasks."""
+    if isinstance(o, dict):
+        return {k: _copy(v) for k, v in o.items()}
+    if isinstance(o, list):
+        return [_copy(v) for v in o]
+    return o
 
 
 class NotFoundError(KeyError):
@@ -86,6 +99,14 @@ class ResourceStore:
         # per-namespace FIFO of event names for TTL-style capping
         self._event_fifo: Dict[str, List[str]] = {}
         self.max_events_per_namespace = 20000
+        # O(1) event aggregation: (ns, involved, type, reason, message) ->
+        # event name.  A linear bucket scan here cost ~tens of seconds per
+        # 1k-task bench wave once the bucket filled (round-1 wave drift).
+        self._event_index: Dict[Tuple, str] = {}
+        # (ns, involved name) -> [event names] — events_for and delete-GC
+        self._events_by_obj: Dict[Tuple[str, str], List[str]] = {}
+        # serialized "put" lines for O(live) compaction without re-dumping
+        self._snap_lines: Dict[Tuple[str, str, str], str] = {}
         self._rv = 0
         self._watches: List[_Watch] = []
         self._wal_path = wal_path
@@ -110,35 +131,39 @@ class ResourceStore:
         with open(path, "rb") as f:
             raw = f.read()
         offset = 0
-        with open(path, "r", encoding="utf-8") as f:
-            for line in f:
-                raw_len = len(line.encode("utf-8"))
-                line = line.strip()
-                if not line:
-                    offset += raw_len
-                    good_end = offset
-                    continue
-                try:
-                    rec = json.loads(line)
-                except json.JSONDecodeError:
-                    offset += raw_len
-                    continue  # torn tail write after a crash
+        # parse from the raw bytes: a non-UTF-8 byte in a torn/corrupt line
+        # must be skipped like bad JSON, not abort recovery (text-mode
+        # iteration would raise UnicodeDecodeError and lose everything)
+        for raw_line in raw.split(b"\n"):
+            raw_len = len(raw_line) + 1  # the split consumed the newline
+            if offset + len(raw_line) >= len(raw):
+                raw_len = len(raw_line)  # last fragment had no newline
+            line_b = raw_line.strip()
+            if not line_b:
                 offset += raw_len
                 good_end = offset
-                op, obj = rec.get("op"), rec.get("obj")
-                if not obj:
-                    continue
-                kind = obj.get("kind")
-                m = obj.get("metadata", {})
-                ns, name = m.get("namespace", "default"), m.get("name")
-                if not kind or not name:
-                    continue
-                bucket = self._data.setdefault(kind, {}).setdefault(ns, {})
-                if op == "delete":
-                    bucket.pop(name, None)
-                else:
-                    bucket[name] = obj
-                self._rv = max(self._rv, int(m.get("resourceVersion", 0)))
+                continue
+            try:
+                rec = json.loads(line_b.decode("utf-8"))
+            except (json.JSONDecodeError, UnicodeDecodeError):
+                offset += raw_len
+                continue  # torn tail write after a crash
+            offset += raw_len
+            good_end = offset
+            op, obj = rec.get("op"), rec.get("obj")
+            if not obj:
+                continue
+            kind = obj.get("kind")
+            m = obj.get("metadata", {})
+            ns, name = m.get("namespace", "default"), m.get("name")
+            if not kind or not name:
+                continue
+            bucket = self._data.setdefault(kind, {}).setdefault(ns, {})
+            if op == "delete":
+                bucket.pop(name, None)
+            else:
+                bucket[name] = obj
+            self._rv = max(self._rv, int(m.get("resourceVersion", 0)))
         # truncate away any torn tail so the next append starts a fresh
         # line (appending onto a torn fragment would corrupt BOTH records)
         if good_end < len(raw):
@@ -154,11 +179,33 @@ class ResourceStore:
                             self._owned_by.setdefault(ref["uid"], []).append(
                                 (kind, ns, name)
                             )
+        # rebuild the event aggregation/involved-object indexes
+        for ns, objs in self._data.get(EVENT, {}).items():
+            ordered = sorted(
+                objs.items(), key=lambda kv: int(kv[1]["metadata"].get("resourceVersion", 0))
+            )
+            fifo = self._event_fifo.setdefault(ns, [])
+            for name, ev in ordered:
+                io = ev.get("involvedObject", {})
+                self._event_index[
+                    (ns, io.get("name"), ev.get("type"), ev.get("reason"), ev.get("message"))
+                ] = name
+                self._events_by_obj.setdefault((ns, io.get("name")), []).append(name)
+                fifo.append(name)
 
     def _append_wal(self, op: str, obj: Dict[str, Any]) -> None:
         if self._wal_file is None:
             return
         line = json.dumps({"op": op, "obj": obj}, separators=(",", ":")) + "\n"
+        key = (
+            obj.get("kind", ""),
+            obj.get("metadata", {}).get("namespace", "default"),
+            obj.get("metadata", {}).get("name", ""),
+        )
+        if op == "put":
+            self._snap_lines[key] = line
+        else:
+            self._snap_lines.pop(key, None)
         self._wal_file.write(line)
         self._wal_written += len(line)
         if self._compact_bytes and self._wal_written > self._compact_bytes:
@@ -175,7 +222,12 @@ class ResourceStore:
                 self._last_fsync = now
 
     def compact(self) -> None:
-        """Rewrite the WAL as one snapshot line per live object."""
+        """Rewrite the WAL as one snapshot line per live object.
+
+        Uses the serialized-line cache where possible: round 1's compaction
+        re-dumped every live object (plus fsync) while holding the store
+        lock — a multi-second stall that showed up as episodic 2x wave-time
+        spikes in the driver bench."""
         if not self._wal_path:
             return
         with self._lock:
@@ -183,11 +235,16 @@ class ResourceStore:
             with open(tmp, "w", encoding="utf-8") as f:
                 for kind, nss in self._data.items():
                     for ns, objs in nss.items():
-                        for obj in objs.values():
-                            f.write(
-                                json.dumps({"op": "put", "obj": obj}, separators=(",", ":"))
-                                + "\n"
-                            )
+                        for name, obj in objs.items():
+                            line = self._snap_lines.get((kind, ns, name))
+                            if line is None:
+                                line = (
+                                    json.dumps(
+                                        {"op": "put", "obj": obj}, separators=(",", ":")
+                                    )
+                                    + "\n"
+                                )
+                            f.write(line)
                 f.flush()
                 os.fsync(f.fileno())
             if self._wal_file:
@@ -209,7 +266,7 @@ class ResourceStore:
                 w.q.put(ev)
 
     def create(self, obj: Dict[str, Any]) -> Dict[str, Any]:
-        obj = copy.deepcopy(obj)
+        obj = _copy(obj)
         kind = obj["kind"]
         m = obj.setdefault("metadata", {})
         ns = m.setdefault("namespace", "default")
@@ -233,14 +290,14 @@ class ResourceStore:
                 if ref.get("uid"):
                     self._owned_by.setdefault(ref["uid"], []).append((kind, ns, name))
             self._append_wal("put", obj)
-            out = copy.deepcopy(obj)
+            out = _copy(obj)
             self._notify(WatchEvent("ADDED", kind, out))
         return out
 
     def get(self, kind: str, name: str, namespace: str = "default") -> Optional[Dict[str, Any]]:
         with self._lock:
             obj = self._data.get(kind, {}).get(namespace, {}).get(name)
-            return copy.deepcopy(obj) if obj is not None else None
+            return _copy(obj) if obj is not None else None
 
     def list(
         self,
@@ -258,7 +315,7 @@ class ResourceStore:
                         labels = obj.get("metadata", {}).get("labels", {}) or {}
                         if any(labels.get(k) != v for k, v in label_selector.items()):
                             continue
-                    out.append(copy.deepcopy(obj))
+                    out.append(_copy(obj))
             return out
 
     def _check_rv(self, current: Dict[str, Any], incoming: Dict[str, Any]) -> None:
@@ -293,13 +350,13 @@ class ResourceStore:
                 # does not bump resourceVersion or emit a watch event —
                 # without this, reconcilers that rewrite unchanged status
                 # self-trigger forever through their own MODIFIED events
-                return copy.deepcopy(cur)
+                return _copy(cur)
             self._rv += 1
             if which == "status":
-                cur["status"] = copy.deepcopy(obj.get("status", {}))
+                cur["status"] = _copy(obj.get("status", {}))
             else:
-                cur["spec"] = copy.deepcopy(obj.get("spec", {}))
-                newm = copy.deepcopy(obj.get("metadata", {}))
+                cur["spec"] = _copy(obj.get("spec", {}))
+                newm = _copy(obj.get("metadata", {}))
                 keep = cur["metadata"]
                 for k in ("labels", "annotations", "ownerReferences", "deletionTimestamp"):
                     if k in newm:
@@ -309,7 +366,7 @@ class ResourceStore:
                 keep["generation"] = int(keep.get("generation", 1)) + 1
             cur["metadata"]["resourceVersion"] = self._rv
             self._append_wal("put", cur)
-            out = copy.deepcopy(cur)
+            out = _copy(cur)
             self._notify(WatchEvent("MODIFIED", kind, out))
         return out
 
@@ -320,12 +377,19 @@ class ResourceStore:
             if obj is None:
                 return False
             self._append_wal("delete", obj)
-            self._notify(WatchEvent("DELETED", kind, copy.deepcopy(obj)))
+            self._notify(WatchEvent("DELETED", kind, _copy(obj)))
             # cascade via the owner index (k8s GC role), O(children)
             uid = obj.get("metadata", {}).get("uid")
             if uid:
                 for (k2, ns2, n2) in self._owned_by.pop(uid, []):
                     self.delete(k2, n2, ns2)
+            # GC the object's events (k8s expires events by TTL; here the
+            # involved object's deletion is the expiry signal — without
+            # this, bench-style create/delete churn pins the event bucket
+            # at its cap and every aggregation lookup degrades)
+            if kind != EVENT:
+                for ev_name in list(self._events_by_obj.get((namespace, name), [])):
+                    self._drop_event(namespace, ev_name)
         return True
 
     # --------------------------------------------------------------- watches
@@ -356,59 +420,87 @@ class ResourceStore:
         (the reference has ~40 recorder.Event call sites, e.g.
         task/state_machine.go:224,628,731)."""
         m = involved.get("metadata", {})
-        # k8s-style event aggregation: an identical (object, reason, message)
-        # event bumps count/lastTimestamp instead of creating a new object
+        ns = m.get("namespace", "default")
+        ikey = (ns, m.get("name"), event_type, reason, message)
         with self._lock:
-            ns_bucket = self._data.get(EVENT, {}).get(m.get("namespace", "default"), {})
-            for ev in ns_bucket.values():
-                if (
-                    ev.get("involvedObject", {}).get("name") == m.get("name")
-                    and ev.get("reason") == reason
-                    and ev.get("message") == message
-                    and ev.get("type") == event_type
-                ):
+            # k8s-style event aggregation: an identical (object, reason,
+            # message) event bumps count/lastTimestamp — O(1) via the index
+            # (a linear bucket scan here cost tens of seconds per 1k-task
+            # bench wave once the bucket filled)
+            ev_name = self._event_index.get(ikey)
+            if ev_name is not None:
+                ev = self._data.get(EVENT, {}).get(ns, {}).get(ev_name)
+                if ev is not None:
                     ev["count"] = int(ev.get("count", 1)) + 1
                     ev["lastTimestamp"] = now_iso()
+                    # durable + visible: rv bump so watchers order it, WAL
+                    # append so a crash does not lose the count
+                    self._rv += 1
+                    ev["metadata"]["resourceVersion"] = self._rv
+                    self._append_wal("put", ev)
+                    self._notify(WatchEvent("MODIFIED", EVENT, _copy(ev)))
                     return
-        name = f'{m.get("name", "obj")}.{self._rv + 1}'
-        ev = {
-            "apiVersion": "v1",
-            "kind": EVENT,
-            "metadata": new_object_meta(name, m.get("namespace", "default")),
-            "spec": {},
-            "status": {},
-            "involvedObject": {
-                "kind": involved.get("kind"),
-                "name": m.get("name"),
-                "namespace": m.get("namespace", "default"),
-                "uid": m.get("uid", ""),
-            },
-            "type": event_type,
-            "reason": reason,
-            "message": message,
-            "count": 1,
-            "lastTimestamp": now_iso(),
-        }
-        with self._lock:
-            ns = m.get("namespace", "default")
+                self._event_index.pop(ikey, None)
+            name = f'{m.get("name", "obj")}.{self._rv + 1}'
+            ev = {
+                "apiVersion": "v1",
+                "kind": EVENT,
+                "metadata": new_object_meta(name, ns),
+                "spec": {},
+                "status": {},
+                "involvedObject": {
+                    "kind": involved.get("kind"),
+                    "name": m.get("name"),
+                    "namespace": ns,
+                    "uid": m.get("uid", ""),
+                },
+                "type": event_type,
+                "reason": reason,
+                "message": message,
+                "count": 1,
+                "lastTimestamp": now_iso(),
+            }
             bucket = self._data.setdefault(EVENT, {}).setdefault(ns, {})
             self._rv += 1
             ev["metadata"]["resourceVersion"] = self._rv
-            bucket[ev["metadata"]["name"]] = ev
+            bucket[name] = ev
+            self._event_index[ikey] = name
+            self._events_by_obj.setdefault((ns, m.get("name")), []).append(name)
             fifo = self._event_fifo.setdefault(ns, [])
-            fifo.append(ev["metadata"]["name"])
+            fifo.append(name)
             # TTL-style cap (k8s events expire after 1h; here: count-bound)
             while len(fifo) > self.max_events_per_namespace:
-                bucket.pop(fifo.pop(0), None)
+                self._drop_event(ns, fifo.pop(0))
             self._append_wal("put", ev)
-            self._notify(WatchEvent("ADDED", EVENT, copy.deepcopy(ev)))
+            self._notify(WatchEvent("ADDED", EVENT, _copy(ev)))
+
+    def _drop_event(self, ns: str, name: str) -> None:
+        """Remove one event + its index entries (lock held)."""
+        ev = self._data.get(EVENT, {}).get(ns, {}).pop(name, None)
+        if ev is None:
+            return
+        io = ev.get("involvedObject", {})
+        self._event_index.pop(
+            (ns, io.get("name"), ev.get("type"), ev.get("reason"), ev.get("message")),
+            None,
+        )
+        bucket = self._events_by_obj.get((ns, io.get("name")))
+        if bucket is not None:
+            try:
+                bucket.remove(name)
+            except ValueError:
+                pass
+            if not bucket:
+                self._events_by_obj.pop((ns, io.get("name")), None)
+        self._snap_lines.pop((EVENT, ns, name), None)
 
     def events_for(self, involved_name: str, namespace: str = "default") -> List[Dict[str, Any]]:
         with self._lock:
+            bucket = self._data.get(EVENT, {}).get(namespace, {})
             out = [
-                copy.deepcopy(e)
-                for e in self._data.get(EVENT, {}).get(namespace, {}).values()
-                if e.get("involvedObject", {}).get("name") == involved_name
+                _copy(bucket[n])
+                for n in self._events_by_obj.get((namespace, involved_name), [])
+                if n in bucket
             ]
             out.sort(key=lambda e: int(e["metadata"].get("resourceVersion", 0)))
             return out

@@ -44,42 +44,58 @@ def make_prompt(rng: random.Random, n_chars: int = 600) -> str:
 
 def run_wave(cp, agent_name: str, concurrency: int, rng: random.Random,
              timeout_s: float = 1200.0):
-    """Create C tasks, wait until all reach FinalAnswer; returns latencies."""
+    """Create C tasks, wait until all reach FinalAnswer; returns latencies.
+
+    Completion is watch-driven: round 1 polled every task's full object
+    every 20 ms, which cost ~50 s of GIL + store-lock time per wave and
+    starved the engine thread — the harness must not be the bottleneck of
+    the thing it measures."""
+    import queue as _queue
+
     from agentcontrolplane_amd.api.types import TASK, TaskPhase, make_resource
 
     t_wave0 = time.monotonic()
+    watch_q = cp.store.watch(kinds={TASK})
     names = []
     t_create = {}
-    for i in range(concurrency):
-        name = f"bench-{rng.randrange(1 << 30):08x}-{i}"
-        cp.store.create(
-            make_resource(
-                TASK,
-                name,
-                spec={"agentRef": {"name": agent_name}, "userMessage": make_prompt(rng)},
+    try:
+        for i in range(concurrency):
+            name = f"bench-{rng.randrange(1 << 30):08x}-{i}"
+            cp.store.create(
+                make_resource(
+                    TASK,
+                    name,
+                    spec={"agentRef": {"name": agent_name}, "userMessage": make_prompt(rng)},
+                )
             )
-        )
-        t_create[name] = time.monotonic()
-        names.append(name)
-    t_created = time.monotonic()
-    deadline = time.monotonic() + timeout_s
-    latencies = {}
-    pending = set(names)
-    while pending:
-        if time.monotonic() > deadline:
-            raise TimeoutError(f"{len(pending)} tasks unfinished (of {concurrency})")
-        done = set()
-        for name in pending:
-            t = cp.store.get(TASK, name)
-            phase = (t or {}).get("status", {}).get("phase")
+            t_create[name] = time.monotonic()
+            names.append(name)
+        t_created = time.monotonic()
+        deadline = time.monotonic() + timeout_s
+        latencies = {}
+        pending = set(names)
+        while pending:
+            try:
+                ev = watch_q.get(timeout=max(0.0, min(5.0, deadline - time.monotonic())))
+            except _queue.Empty:
+                if time.monotonic() > deadline:
+                    raise TimeoutError(
+                        f"{len(pending)} tasks unfinished (of {concurrency})"
+                    ) from None
+                continue
+            name = ev.obj.get("metadata", {}).get("name")
+            if name not in pending:
+                continue
+            phase = ev.obj.get("status", {}).get("phase")
             if phase == TaskPhase.FINAL_ANSWER:
                 latencies[name] = time.monotonic() - t_create[name]
-                done.add(name)
+                pending.discard(name)
             elif phase == TaskPhase.FAILED:
-                raise RuntimeError(f"task {name} failed: {t['status'].get('error')}")
-        pending -= done
-        if pending:
-            time.sleep(0.02)
+                raise RuntimeError(
+                    f"task {name} failed: {ev.obj['status'].get('error')}"
+                )
+    finally:
+        cp.store.stop_watch(watch_q)
     # forensics: the wave is gated by its slowest task — dump its event
     # timeline (stderr; the stdout JSON line stays clean)
     t_polled = time.monotonic()

@@ -197,7 +197,9 @@ def test_openai_chat_completions_endpoint():
         )
         assert r.status_code == 200, r.text
         body = r.json()
-        assert body["choices"][0]["finish_reason"] == "stop"
+        # random-init weights rarely emit EOT within 8 tokens: 'length'
+        # must pass through unmapped (truncation visible to OpenAI clients)
+        assert body["choices"][0]["finish_reason"] in ("stop", "length")
         assert body["usage"]["completion_tokens"] <= 8
         # tool-calling turn
         r = client.post(
