@@ -327,6 +327,12 @@ class TaskReconciler(Reconciler):
         status = task["status"]
         spec = task.get("spec", {})
         key = (ns, name)
+        labels = task["metadata"].get("labels", {}) or {}
+        if labels.get(V1BETA3_LABEL) == "true" and self._v1beta3_parked(status):
+            if status.get("statusDetail") != "Awaiting next inbound event":
+                status["statusDetail"] = "Awaiting next inbound event"
+                self.store.update_status(task)
+            return Result()  # parked: a new inbound event opens a new Task
         marker = len(status.get("contextWindow", []))
 
         # a completed async turn? process it under the lock/lease
@@ -463,7 +469,11 @@ class TaskReconciler(Reconciler):
         labels = task["metadata"].get("labels", {}) or {}
         prev_phase = status.get("phase", "")
         if output.content:
-            if labels.get(V1BETA3_LABEL) == "true" and not self._v1beta3_round_done(status):
+            if labels.get(V1BETA3_LABEL) == "true":
+                # reference parity: EVERY content output on a v1beta3 task
+                # becomes a respond_to_human ToolCall (state_machine.go:
+                # 609-611 has no round-done special case); the loop back to
+                # ReadyForLLM then parks (see _send_llm_request_locked)
                 return self._v1beta3_final_answer(task, output)
             status.update(
                 {
@@ -623,19 +633,27 @@ class TaskReconciler(Reconciler):
     # -------------------------------------------------------------- v1beta3
 
     @staticmethod
-    def _v1beta3_round_done(status) -> bool:
-        """True once this thread turn already delivered a respond_to_human
-        round.  The reference loops a v1beta3 task back to ReadyForLLM after
-        the respond_to_human ToolCall succeeds (checkToolCalls has no special
-        case); the conversation then parks until the next inbound event.  We
-        terminate the turn instead — content produced after a completed
-        respond_to_human round is a plain FinalAnswer — which keeps the flow
-        finite under scripted/mock LLMs and identical up to that round."""
-        for m in status.get("contextWindow", []):
-            if m.get("role") == "assistant":
-                for tc in m.get("toolCalls", []) or []:
-                    if tc.get("function", {}).get("name") == "respond_to_human":
-                        return True
+    def _v1beta3_parked(status) -> bool:
+        """True when the v1beta3 thread's last completed action was a
+        delivered respond_to_human round: the reference loops the task back
+        to ReadyForLLM after the ToolCall succeeds (checkToolCalls has no
+        special case) and the conversation effectively waits for the next
+        inbound event — each new event arrives as a NEW Task carrying the
+        same threadID (server.go:1384-1545), so calling the LLM again here
+        would just re-answer its own delivery receipt.  Parking preserves
+        the reference's phase sequence through the respond_to_human round
+        while keeping the loop finite."""
+        cw = status.get("contextWindow", [])
+        if not cw or cw[-1].get("role") != "tool":
+            return False
+        last_tc_id = cw[-1].get("toolCallId", "")
+        for m in reversed(cw):
+            if m.get("role") != "assistant":
+                continue
+            for tc in m.get("toolCalls", []) or []:
+                if tc.get("id") == last_tc_id:
+                    return tc.get("function", {}).get("name") == "respond_to_human"
+            return False
         return False
 
     def _v1beta3_final_answer(self, task, output: Message) -> Result:
@@ -681,7 +699,7 @@ class TaskReconciler(Reconciler):
         self.store.update_status(task)
         name = task["metadata"]["name"]
         ns = task["metadata"].get("namespace", "default")
-        new_name = f"{name}-{req_id}-tc-01"
+        new_name = f"{name}-{req_id}-respond-to-human"  # state_machine.go:1022
         self.store.create(
             {
                 "apiVersion": "acp.humanlayer.dev/v1alpha1",
@@ -689,7 +707,12 @@ class TaskReconciler(Reconciler):
                 "metadata": {
                     "name": new_name,
                     "namespace": ns,
-                    "labels": {LABEL_TASK: name, LABEL_TCREQ: req_id},
+                    "labels": {
+                        LABEL_TASK: name,
+                        LABEL_TCREQ: req_id,
+                        V1BETA3_LABEL: "true",
+                        "acp.humanlayer.dev/tool-type": "respond_to_human",
+                    },
                     "ownerReferences": [owner_ref(task)],
                 },
                 "spec": {

@@ -392,7 +392,9 @@ class ToolCallReconciler(Reconciler):
                 client.notify_final_result(message)
             except Exception as e:
                 return self._fail(tc, str(e))
-            return self._finish(tc, message)
+            # executor.go:400 result format
+            call_id = tc["spec"].get("toolCallId", "")
+            return self._finish(tc, f"Response sent to human, call ID: {call_id}")
         try:
             client = self.humanlayer.new_client(
                 namespace=ns, run_id=tc["metadata"]["name"], channel=channel_spec

@@ -5,10 +5,12 @@ providers are:
 
 - ``mock``   deterministic scripted client (tests, config-1 bench)
 - ``local``  the in-process MI355X inference engine (the whole point)
-- openai/anthropic/mistral/google/vertex — recognized for spec parity; they
-  validate config shape but cannot reach the network in this environment, so
-  their ``send_request`` raises a 502-class LLMRequestError unless a
-  ``transport`` is injected (tests inject an httpx mock).
+- openai/anthropic/mistral/google/vertex — real httpx clients with the
+  reference's message/tool conversion rules (llmclient/remote.py); there is
+  no egress in this deployment, so production use targets
+  ``parameters.baseUrl`` (gateway/self-hosted endpoints) and tests run
+  against an in-process mock server exactly as the reference's e2e does
+  (test_getting_started.go:250-261).
 """
 from __future__ import annotations
 
@@ -19,39 +21,25 @@ from .base import LLMClient, LLMRequestError
 KNOWN_PROVIDERS = ("openai", "anthropic", "mistral", "google", "vertex", "mock", "local")
 
 
-class _RemoteStubClient(LLMClient):
-    def __init__(self, provider: str, parameters: Dict[str, Any], api_key: str,
-                 transport: Optional[Callable] = None):
-        self.provider = provider
-        self.parameters = parameters
-        self.api_key = api_key
-        self.transport = transport
-
-    def send_request(self, messages, tools):
-        if self.transport is not None:
-            return self.transport(self.provider, self.parameters, messages, tools)
-        raise LLMRequestError(
-            502, f"remote provider {self.provider!r} is unreachable in this deployment"
-        )
-
-
 class LLMClientFactory:
     """create_client(llm_resource, api_key) -> LLMClient.
 
     ``engine_provider`` is a callable returning the shared engine-backed
     client (injected by the process wiring so one engine serves every
-    reconciler worker); ``mock_factory`` can be overridden by tests.
+    reconciler worker); ``mock_factory`` can be overridden by tests;
+    ``http_transport`` (an httpx transport) lets tests intercept the remote
+    providers without sockets.
     """
 
     def __init__(
         self,
         engine_provider: Optional[Callable[[Dict[str, Any]], LLMClient]] = None,
         mock_factory: Optional[Callable[[Dict[str, Any]], LLMClient]] = None,
-        remote_transport: Optional[Callable] = None,
+        http_transport=None,
     ):
         self._engine_provider = engine_provider
         self._mock_factory = mock_factory
-        self._remote_transport = remote_transport
+        self._http_transport = http_transport
 
     def create_client(self, llm: Dict[str, Any], api_key: str = "") -> LLMClient:
         provider = llm.get("spec", {}).get("provider", "")
@@ -69,9 +57,8 @@ class LLMClientFactory:
                     503, "local provider requested but no engine is attached to this manager"
                 )
             return self._engine_provider(llm)
-        return _RemoteStubClient(
-            provider,
-            llm.get("spec", {}).get("parameters", {}) or {},
-            api_key,
-            transport=self._remote_transport,
+        from .remote import create_remote_client
+
+        return create_remote_client(
+            provider, llm.get("spec", {}) or {}, api_key, transport=self._http_transport
         )

@@ -129,10 +129,18 @@ def test_v1beta3_event_flow(client, cp):
     )
     assert r.status_code == 201
     task_name = r.json()["taskName"]
+    # reference parity (state_machine.go:609-611, 968-1067): the content
+    # turn becomes a respond_to_human ToolCall, checkToolCalls loops the
+    # task back to ReadyForLLM, and the thread parks awaiting the next
+    # inbound event (which would arrive as a NEW Task on the same threadID)
     task = wait_for(
-        lambda: (cp.store.get(TASK, task_name) or {}).get("status", {}).get("phase")
-        == TaskPhase.FINAL_ANSWER
-        and cp.store.get(TASK, task_name),
+        lambda: (
+            (cp.store.get(TASK, task_name) or {}).get("status", {}).get("phase")
+            == TaskPhase.READY_FOR_LLM
+            and (cp.store.get(TASK, task_name) or {})["status"].get("statusDetail")
+            == "Awaiting next inbound event"
+            and cp.store.get(TASK, task_name)
+        ),
         timeout=30,
     )
     # final answer flowed through a respond_to_human tool call
