@@ -18,9 +18,23 @@ from .manager import Reconciler, Result
 
 
 def default_verify_api_key(api_key: str, channel_id: str = "") -> Dict[str, str]:
-    """Stand-in for GET humanlayer/v1/project (contactchannel_controller.go:36).
+    """Key verification against the HumanLayer API when an endpoint is
+    configured, offline stand-in otherwise.
 
-    Returns project/org slugs, raises on an invalid key."""
+    The reference GETs ``humanLayerAPIURL`` (a package var defaulting to
+    the live SaaS, contactchannel_controller.go:36) and, for
+    channel-specific auth, the channel endpoint
+    (state_machine.go:173-252).  Here: with ``HUMANLAYER_API_BASE`` set
+    the same HTTP verification runs (humanlayer/wire.py,
+    test-overridable exactly like the reference's httptest pattern);
+    without it — this deployment has no egress — a prefix check stands
+    in so channels can exist offline."""
+    import os
+
+    if os.environ.get("HUMANLAYER_API_BASE"):
+        from ..humanlayer.wire import verify_api_key_http
+
+        return verify_api_key_http(api_key, channel_id)
     if api_key.startswith(("sk-", "hl-")):
         return {"projectSlug": "local-project", "orgSlug": "local-org"}
     raise PermissionError("invalid HumanLayer API key")

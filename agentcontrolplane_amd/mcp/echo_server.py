@@ -10,6 +10,9 @@ from __future__ import annotations
 
 import json
 import sys
+import threading
+
+_OUT_LOCK = threading.Lock()
 
 TOOLS = [
     {
@@ -27,6 +30,11 @@ TOOLS = [
         "inputSchema": {"type": "object", "properties": {"text": {"type": "string"}}},
     },
     {"name": "noop", "description": "No-op", "inputSchema": {"type": "object", "properties": {}}},
+    {
+        "name": "sleep",
+        "description": "Sleep for `seconds`, then return 'slept' (tests the client's concurrent in-flight requests)",
+        "inputSchema": {"type": "object", "properties": {"seconds": {"type": "number"}}},
+    },
 ]
 
 
@@ -48,10 +56,30 @@ def handle(method, params):
             text = args.get("text", "")
         elif name == "noop":
             text = "ok"
+        elif name == "sleep":
+            import time
+
+            time.sleep(float(args.get("seconds", 0)))
+            text = "slept"
         else:
             return {"isError": True, "content": [{"type": "text", "text": f"unknown tool {name}"}]}
         return {"content": [{"type": "text", "text": text}]}
     raise KeyError(method)
+
+
+def _respond(msg) -> None:
+    try:
+        result = handle(msg.get("method", ""), msg.get("params", {}) or {})
+        resp = {"jsonrpc": "2.0", "id": msg["id"], "result": result}
+    except KeyError:
+        resp = {
+            "jsonrpc": "2.0",
+            "id": msg["id"],
+            "error": {"code": -32601, "message": "method not found"},
+        }
+    with _OUT_LOCK:
+        sys.stdout.write(json.dumps(resp) + "\n")
+        sys.stdout.flush()
 
 
 def main() -> None:
@@ -65,17 +93,9 @@ def main() -> None:
             continue
         if "id" not in msg:
             continue  # notification
-        try:
-            result = handle(msg.get("method", ""), msg.get("params", {}) or {})
-            resp = {"jsonrpc": "2.0", "id": msg["id"], "result": result}
-        except KeyError:
-            resp = {
-                "jsonrpc": "2.0",
-                "id": msg["id"],
-                "error": {"code": -32601, "message": "method not found"},
-            }
-        sys.stdout.write(json.dumps(resp) + "\n")
-        sys.stdout.flush()
+        # thread-per-request: a slow tool must not block other callers
+        # (exercises the client's concurrent in-flight request ids)
+        threading.Thread(target=_respond, args=(msg,), daemon=True).start()
 
 
 if __name__ == "__main__":

@@ -26,7 +26,7 @@ def test_stdio_connect_list_call(store):
     mgr = MCPServerManager(store)
     try:
         tools = mgr.connect_server(_stdio_server_obj())
-        assert {t["name"] for t in tools} == {"add", "echo", "noop"}
+        assert {t["name"] for t in tools} == {"add", "echo", "noop", "sleep"}
         assert mgr.call_tool("calc", "add", {"a": 2, "b": 3}) == "5.0"
         assert mgr.call_tool("calc", "echo", {"text": "hi"}) == "hi"
         server, tool = mgr.find_server_for_tool("calc__add")
