@@ -42,12 +42,22 @@ def cmd_serve(args) -> None:
                 num_kv_blocks=args.kv_blocks,
             )
         )
+    store = None
+    if getattr(args, "kube", False):
+        # apiserver-backed store: in-cluster credentials or --kube-url
+        from .store.kube import KubeStore
+
+        store = KubeStore(base_url=args.kube_url or None)
+    import os as _os
+
     cp = ControlPlane(
         wal_path=args.wal,
         engine=engine,
         auto_approve="approve" if args.auto_approve else None,
+        store=store,
+        pod_name=_os.environ.get("POD_NAME", "acp-controller-0"),
     )
-    cp.start()
+    cp.start(leader_elect=getattr(args, "leader_elect", False))
     app = cp.rest_app
     from .server.admin import add_admin_routes
 
@@ -189,6 +199,12 @@ def main() -> None:
     s.add_argument("--device", default=None, help="cuda | cpu | none (no engine)")
     s.add_argument("--kv-blocks", type=int, default=None)
     s.add_argument("--auto-approve", action="store_true")
+    s.add_argument("--kube", action="store_true",
+                   help="back the store with the Kubernetes apiserver "
+                   "(in-cluster credentials or --kube-url)")
+    s.add_argument("--kube-url", default=None)
+    s.add_argument("--leader-elect", action="store_true",
+                   help="multi-replica active/passive via a coordination Lease")
     s.set_defaults(fn=cmd_serve)
 
     a = sub.add_parser("apply", help="apply CRD-style YAML manifests")

@@ -44,6 +44,7 @@ class ControlPlane:
         # backing store (the reference's multi-pod deployment over one etcd)
         self.store = store if store is not None else ResourceStore(wal_path=wal_path, fsync=fsync)
         self._owns_store = store is None
+        self.pod_name = pod_name
         self.tracer = get_tracer()
         self.engine = engine
         self.mcp = MCPServerManager(self.store)
@@ -90,14 +91,62 @@ class ControlPlane:
 
     # ------------------------------------------------------------- lifecycle
 
-    def start(self) -> "ControlPlane":
-        self.manager.start()
+    #: leader-election lease constants (cmd/main.go:208-226: election ID
+    #: 2994eb7c.humanlayer.dev; controller-runtime defaults 15s/10s/2s)
+    LEADER_LEASE = "2994eb7c.humanlayer.dev"
+    LEADER_LEASE_DURATION = 15.0
+    LEADER_RETRY = 2.0
+
+    def start(self, leader_elect: bool = False) -> "ControlPlane":
+        """With ``leader_elect`` the reconcilers start only once this
+        replica holds the leader Lease, and stop if leadership is lost —
+        multiple pods over one store run active/passive like the
+        reference's manager (cmd/main.go:208-226).  The REST server stays
+        up on every replica (reads work anywhere; the reference runs its
+        API leader-only, runnable.go:30-33 — here followers serve reads
+        and writes land in the shared store for the leader to reconcile)."""
+        if not leader_elect:
+            self.manager.start()
+            return self
+        self.is_leader = False
+        self._elect_stop = False
+        import threading
+
+        def _elect_loop() -> None:
+            holder = self.pod_name
+            while not self._elect_stop:
+                got = self.store.acquire_lease(
+                    self.LEADER_LEASE, holder, self.LEADER_LEASE_DURATION
+                )
+                if got and not self.is_leader:
+                    self.is_leader = True
+                    self.manager.start()
+                elif not got and self.is_leader:
+                    # leadership lost: stop reconciling immediately
+                    self.is_leader = False
+                    self.manager.stop()
+                import time as _t
+
+                _t.sleep(
+                    self.LEADER_LEASE_DURATION / 3 if self.is_leader else self.LEADER_RETRY
+                )
+
+        self._elect_thread = threading.Thread(
+            target=_elect_loop, name="acp-leader-elect", daemon=True
+        )
+        self._elect_thread.start()
         return self
 
     def stop(self) -> None:
+        if getattr(self, "_elect_thread", None) is not None:
+            self._elect_stop = True
+            self._elect_thread.join(timeout=5)
+            self._elect_thread = None
+            if getattr(self, "is_leader", False):
+                self.store.release_lease(self.LEADER_LEASE, self.pod_name)
         self.manager.stop()
         self.mcp.close()
-        if self._owns_store:
+        if self._owns_store and hasattr(self.store, "close"):
             self.store.close()
 
     def __enter__(self) -> "ControlPlane":
