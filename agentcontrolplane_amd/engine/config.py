@@ -135,6 +135,9 @@ class EngineConfig:
     request_timeout_s: float = 600.0
     # HF-named safetensors dir; None = random-init (no network here)
     checkpoint_path: Optional[str] = None
+    # HF tokenizer.json; defaults to <checkpoint_path>/tokenizer.json when
+    # present, else the byte-level tokenizer (synthetic workloads)
+    tokenizer_path: Optional[str] = None
 
     def model_config(self) -> ModelConfig:
         if self.model not in PRESETS:

@@ -35,7 +35,21 @@ class InferenceEngine:
         self.device = torch.device(config.device)
         if self.device.type == "cuda":
             self._load_gemm_tunings()
-        self.tokenizer = ByteTokenizer(self.mcfg.vocab_size)
+        import os as _os
+
+        tok_path = config.tokenizer_path
+        if tok_path is None and config.checkpoint_path:
+            cand = _os.path.join(config.checkpoint_path, "tokenizer.json")
+            if _os.path.exists(cand):
+                tok_path = cand
+        if tok_path:
+            from .tokenizer import HFTokenizer
+
+            self.tokenizer = HFTokenizer(tok_path, self.mcfg.vocab_size)
+        else:
+            self.tokenizer = ByteTokenizer(self.mcfg.vocab_size)
+        self._eot = self.tokenizer.eot
+        self._live = min(self.tokenizer.live_vocab, self.mcfg.vocab_size)
         self.tp_world = config.tensor_parallel
         self.tp_rank = 0
         if model is not None:
@@ -64,6 +78,22 @@ class InferenceEngine:
             num_blocks, config.kv_block_size, prefer_native=True
         )
         self.scheduler = Scheduler(config, self.bm, self.device)
+        if getattr(self.tokenizer, "token_bytes", None) and self._live > N_SPECIAL:
+            # BPE tokenizer: token-trie constrained decoding (token_grammar)
+            from .token_grammar import TokenGrammar, TokenTrie
+
+            tb = [self.tokenizer.token_bytes(i) for i in range(self._live)]
+            trie = TokenTrie(tb)
+            mask_cache: Dict = {}
+
+            def _grammar_factory(request):
+                max_args = min(2048, max(32, request.sampling.max_tokens * 4))
+                return TokenGrammar(
+                    trie, tb, self._eot, request.tools, max_args_len=max_args,
+                    pre_in_prompt=request.pre_in_prompt, mask_cache=mask_cache,
+                )
+
+            self.scheduler.grammar_factory = _grammar_factory
         self.graph_runner = None
         # MoE decode batches under dense_moe_threshold take the all-experts
         # dense path (static control flow) and are capture-safe; larger MoE
@@ -112,7 +142,18 @@ class InferenceEngine:
         self._ps_temp = stager.persistent(cap, (), _np.float32)
         self._ps_topk = stager.persistent(cap, (), _np.int64)
         self._ps_topp = stager.persistent(cap, (), _np.float32)
-        self._ps_mask = stager.persistent(cap, (N_SPECIAL,), _np.bool_)
+        if self._live <= 512:
+            # byte vocab: dense per-row masks (bitonic sampler)
+            self._ps_mask = stager.persistent(cap, (self._live,), _np.bool_)
+            self._ps_maskmap = None
+            self._mask_rows = cap
+        else:
+            # BPE vocab: compact mask rows + per-row indirection (a dense
+            # [B, 128k] pinned mask would be hundreds of MB); rows beyond
+            # the cap take a slow in-place masked_fill path
+            self._mask_rows = 128
+            self._ps_mask = stager.persistent(self._mask_rows, (self._live,), _np.bool_)
+            self._ps_maskmap = stager.persistent(cap, (), _np.int32)
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
         self._pending: List[InferenceRequest] = []
@@ -264,7 +305,9 @@ class InferenceEngine:
             # masked decode steps into one chunked-prefill extension
             from .grammar import ToolCallGrammar
 
-            prompt_ids = prompt_ids + list(ToolCallGrammar.PRE)
+            prompt_ids = prompt_ids + self.tokenizer.encode_text(
+                ToolCallGrammar.PRE.decode()
+            )
             pre_in_prompt = True
         return InferenceRequest(prompt_ids, sampling, constrained, tools, pre_in_prompt)
 
@@ -332,8 +375,9 @@ class InferenceEngine:
                     raise TimeoutError(f"request {req.request_id} timed out") from None
                 if tok is None:
                     break
-                if tok < 256:
-                    buf.append(tok)
+                tb = self.tokenizer.token_bytes(tok)
+                if tb:
+                    buf.extend(tb)
                     try:
                         text = buf.decode("utf-8")
                     except UnicodeDecodeError:
@@ -519,7 +563,7 @@ class InferenceEngine:
         if not seqs:
             return {"batch": batch, "seqs": [], "tokens": None}
         B = len(seqs)
-        live = logits[:, :N_SPECIAL]  # sampling restricted to decodable ids
+        live = logits[:, :self._live]  # sampling restricted to decodable ids
         self._m["ls_gather_s"] = self._m.get("ls_gather_s", 0.0) + time.monotonic() - ta
         tb = time.monotonic()
         # per-sequence sampling params are static: refill the persistent
@@ -542,19 +586,55 @@ class InferenceEngine:
             )
         temps, top_ks, top_ps = self._sampling_cache
         mask = None
+        mask_map = None
         if any(s is not None and s.grammar is not None for s in seqs):
             tm = time.monotonic()
-            m = self._ps_mask.host()
-            m[:B] = True
-            for i, s in enumerate(seqs):
-                if s is not None and s.grammar is not None:
+            if self._ps_maskmap is None:
+                # dense path (byte vocab)
+                m = self._ps_mask.host()
+                m[:B] = True
+                for i, s in enumerate(seqs):
+                    if s is not None and s.grammar is not None:
+                        allowed = s.grammar.allowed_tokens()
+                        if s.grammar.accepting:
+                            allowed = set(allowed) | {self._eot}
+                        row = m[i]
+                        row[:] = False
+                        row[list(allowed)] = True
+                mask = self._ps_mask.commit(B)
+            else:
+                # compact path (BPE vocab): one mask row per constrained seq
+                m = self._ps_mask.host()
+                mm = self._ps_maskmap.host()
+                mm[:B] = -1
+                j = 0
+                overflow = []
+                for i, s in enumerate(seqs):
+                    if s is None or s.grammar is None:
+                        continue
                     allowed = s.grammar.allowed_tokens()
                     if s.grammar.accepting:
-                        allowed = set(allowed) | {EOT}
-                    row = m[i]
-                    row[:] = False
-                    row[list(allowed)] = True
-            mask = self._ps_mask.commit(B)
+                        allowed = set(allowed) | {self._eot}
+                    if j < self._mask_rows:
+                        row = m[j]
+                        row[:] = False
+                        import numpy as _np2
+
+                        row[_np2.fromiter(allowed, dtype=_np2.int64, count=len(allowed))] = True
+                        mm[i] = j
+                        j += 1
+                    else:
+                        overflow.append((i, allowed))
+                mask = self._ps_mask.commit(max(1, j))
+                mask_map = self._ps_maskmap.commit(B)
+                for i, allowed in overflow:
+                    # slow path beyond the compact-row cap: mask in place
+                    ids = torch.tensor(sorted(allowed), dtype=torch.long,
+                                       device=live.device)
+                    rowmask = torch.zeros(self._live, dtype=torch.bool,
+                                          device=live.device)
+                    rowmask[ids] = True
+                    live[i] = live[i].masked_fill(~rowmask, float("-inf"))
             self._m["mask_time_s"] = (
                 self._m.get("mask_time_s", 0.0) + time.monotonic() - tm
             )
@@ -569,20 +649,45 @@ class InferenceEngine:
         ):
             import numpy as np
 
-            pen = np.zeros((B, N_SPECIAL), dtype=np.float32)
-            for i, s in enumerate(seqs):
-                if s is None or s.state == "finished":
-                    continue
-                sp2 = s.request.sampling
-                if not (sp2.frequency_penalty or sp2.presence_penalty):
-                    continue
-                counts: Dict[int, int] = {}
-                for t in s.request.output_ids:
-                    if t < N_SPECIAL:
-                        counts[t] = counts.get(t, 0) + 1
-                for t, c in counts.items():
-                    pen[i, t] = sp2.frequency_penalty * c + sp2.presence_penalty
-            live = live - self.scheduler.stager.tensor("pen", pen, "float32")
+            if self._live <= 512:
+                pen = np.zeros((B, self._live), dtype=np.float32)
+                for i, s in enumerate(seqs):
+                    if s is None or s.state == "finished":
+                        continue
+                    sp2 = s.request.sampling
+                    if not (sp2.frequency_penalty or sp2.presence_penalty):
+                        continue
+                    counts: Dict[int, int] = {}
+                    for t in s.request.output_ids:
+                        if t < self._live:
+                            counts[t] = counts.get(t, 0) + 1
+                    for t, c in counts.items():
+                        pen[i, t] = sp2.frequency_penalty * c + sp2.presence_penalty
+                live = live - self.scheduler.stager.tensor("pen", pen, "float32")
+            else:
+                # BPE vocab: sparse in-place scatter (a dense [B, 128k]
+                # penalty tensor would be hundreds of MB per step)
+                rows, cols, vals = [], [], []
+                for i, s in enumerate(seqs):
+                    if s is None or s.state == "finished":
+                        continue
+                    sp2 = s.request.sampling
+                    if not (sp2.frequency_penalty or sp2.presence_penalty):
+                        continue
+                    counts = {}
+                    for t in s.request.output_ids:
+                        if t < self._live:
+                            counts[t] = counts.get(t, 0) + 1
+                    for t, c in counts.items():
+                        rows.append(i)
+                        cols.append(t)
+                        vals.append(sp2.frequency_penalty * c + sp2.presence_penalty)
+                if rows:
+                    ri = torch.tensor(rows, dtype=torch.long, device=live.device)
+                    ci = torch.tensor(cols, dtype=torch.long, device=live.device)
+                    vi = torch.tensor(vals, dtype=torch.float32, device=live.device)
+                    live = live.clone() if live._base is not None else live
+                    live[ri, ci] -= vi.to(live.dtype)
         # per-request seeded draws: each seeded request consumes exactly one
         # uniform from its own generator per committed token
         uniforms = None
@@ -600,7 +705,9 @@ class InferenceEngine:
                     g.manual_seed(int(s.request.sampling.seed))
                     s.request.gen = g
                 uniforms[i] = torch.rand(1, device=logits.device, generator=s.request.gen)[0]
-        tokens = ops.softmax_sample(live, temps, top_ks, top_ps, self._gen, mask, uniforms)
+        tokens = ops.softmax_sample(
+            live, temps, top_ks, top_ps, self._gen, mask, uniforms, mask_map
+        )
         self._m["ls_kernel_s"] = self._m.get("ls_kernel_s", 0.0) + time.monotonic() - tc
         batch._tokens_gpu = tokens  # speculative successors gather from this
         return {"batch": batch, "seqs": seqs, "tokens": tokens}
@@ -619,8 +726,8 @@ class InferenceEngine:
             if s.request.first_token_time is None:
                 s.request.first_token_time = time.monotonic()
             if s.grammar is not None:
-                if tok == EOT and s.grammar.accepting:
-                    s.grammar.advance(EOT)
+                if tok == self._eot and s.grammar.accepting:
+                    s.grammar.advance(self._eot)
                     s.request.output_ids.pop()  # EOT is not part of the JSON
                     s.output_ids.pop()
                     self._m["requests_completed"] += 1
@@ -635,7 +742,7 @@ class InferenceEngine:
                     self._m["requests_completed"] += 1
                     self.scheduler.finish_seq(s, "length")
                 continue
-            if tok == EOT:
+            if tok == self._eot:
                 s.request.output_ids.pop()
                 s.output_ids.pop()
                 self._m["requests_completed"] += 1

@@ -27,6 +27,43 @@ HEX = b"0123456789abcdefABCDEF"
 # printable ASCII minus '"' and '\': generation stays single-byte-valid UTF-8
 # (multi-byte sequences would need a UTF-8 sub-automaton to stay decodable)
 STRING_SAFE = bytes(b for b in range(0x20, 0x7F) if b not in (0x22, 0x5C))
+# UTF-8 multi-byte sequences inside generated strings: lead bytes by
+# continuation count (conservative well-formed ranges) + continuations
+UTF8_LEAD = {1: bytes(range(0xC2, 0xE0)), 2: bytes(range(0xE0, 0xF0)),
+             3: bytes(range(0xF0, 0xF5))}
+UTF8_CONT = bytes(range(0x80, 0xC0))
+UTF8_LEADS = UTF8_LEAD[1] + UTF8_LEAD[2] + UTF8_LEAD[3]
+
+
+def _utf8_cont_count(b: int) -> int:
+    if 0xC2 <= b < 0xE0:
+        return 1
+    if 0xE0 <= b < 0xF0:
+        return 2
+    if 0xF0 <= b < 0xF5:
+        return 3
+    return 0
+
+
+def _utf8_first_range(lead: int):
+    """(n_continuations, first_lo, first_hi): the FIRST continuation range
+    depends on the lead byte (overlong/surrogate/out-of-range encodings
+    are illegal UTF-8 and must never be generated)."""
+    if 0xC2 <= lead < 0xE0:
+        return 1, 0x80, 0xBF
+    if lead == 0xE0:
+        return 2, 0xA0, 0xBF
+    if lead == 0xED:
+        return 2, 0x80, 0x9F
+    if 0xE0 <= lead < 0xF0:
+        return 2, 0x80, 0xBF
+    if lead == 0xF0:
+        return 3, 0x90, 0xBF
+    if lead == 0xF4:
+        return 3, 0x80, 0x8F
+    if 0xF0 <= lead < 0xF5:
+        return 3, 0x80, 0xBF
+    return 0, 0, 0
 
 
 def _bs(*parts: Iterable[int]) -> Set[int]:
@@ -62,6 +99,17 @@ class JsonValueMachine:
     #   arr_next     expecting ',' or ']'
     # stack frames: 'O' (in object, after value), 'A' (in array, after value)
 
+    def clone(self) -> "JsonValueMachine":
+        c = object.__new__(JsonValueMachine)
+        c.stack = list(self.stack)
+        c.state = self.state
+        c.max_depth = self.max_depth
+        c.max_len = self.max_len
+        c.count = self.count
+        c.done = self.done
+        c.root_object = self.root_object
+        return c
+
     def allowed(self) -> Set[int]:
         if self.count >= self.max_len:
             return self._closing_allowed()
@@ -75,7 +123,10 @@ class JsonValueMachine:
                 opts |= _bs(b"{[")
             return opts
         if s == "string":
-            return _bs(STRING_SAFE, b'"', b"\\")
+            return _bs(STRING_SAFE, b'"', b"\\", UTF8_LEADS)
+        if s.startswith("str_c"):
+            _, lo, hi = s.split(":")
+            return _bs(range(int(lo), int(hi) + 1))
         if s == "str_escape":
             return _bs(b'"\\/bfnrtu')
         if s.startswith("str_u"):
@@ -217,6 +268,10 @@ class JsonValueMachine:
             else:
                 self.state = "string"
             return
+        if s.startswith("str_c"):
+            n = int(s.split(":")[0][5:])
+            self.state = "string" if n == 1 else f"str_c{n - 1}:128:191"
+            return
         if s.startswith("str_u"):
             n = int(s[5:])
             self.state = "string" if n == 1 else f"str_u{n - 1}"
@@ -340,9 +395,25 @@ class SchemaArgsMachine:
         self.off = 0      # offset within a literal / value state
         self.sub: Optional[JsonValueMachine] = None
         self._bool_rest = b""
+        self.u8 = 0  # pending UTF-8 continuation bytes in a string value
+        self.u8lo, self.u8hi = 0x80, 0xBF
 
     def _cur(self):
         return self.prog[self.pi] if self.pi < len(self.prog) else None
+
+    def clone(self) -> "SchemaArgsMachine":
+        c = object.__new__(SchemaArgsMachine)
+        c.fields = self.fields      # immutable after build
+        c.max_len = self.max_len
+        c.count = self.count
+        c.done = self.done
+        c.prog = self.prog          # immutable after build
+        c.pi = self.pi
+        c.off = self.off
+        c.sub = self.sub.clone() if self.sub is not None else None
+        c._bool_rest = self._bool_rest
+        c.u8, c.u8lo, c.u8hi = self.u8, self.u8lo, self.u8hi
+        return c
 
     def allowed(self) -> Set[int]:
         cur = self._cur()
@@ -353,11 +424,16 @@ class SchemaArgsMachine:
             return {cur[1][self.off]}
         budget_left = self.count < self.max_len
         if kind == "str":
+            if self.u8:
+                return _bs(range(self.u8lo, self.u8hi + 1))
             if self.off == 0:
                 return {0x22}
             opts = {0x22}
             if budget_left:
                 opts |= set(STRING_SAFE)
+                # a multi-byte codepoint needs budget for its continuations
+                if self.count + 4 < self.max_len:
+                    opts |= set(UTF8_LEADS)
             return opts
         if kind == "num":
             if self.off == 0:
@@ -401,6 +477,16 @@ class SchemaArgsMachine:
                     self.done = True
             return
         if kind == "str":
+            if self.u8:
+                self.u8 -= 1
+                self.u8lo, self.u8hi = 0x80, 0xBF
+                self.off += 1
+                return
+            n, lo, hi = _utf8_first_range(b)
+            if n:
+                self.u8, self.u8lo, self.u8hi = n, lo, hi
+                self.off += 1
+                return
             if self.off > 0 and b == 0x22:
                 self.pi += 1
                 self.off = 0
@@ -496,6 +582,20 @@ class ToolCallGrammar:
             self.args = SchemaArgsMachine(schema, max_len=self.max_args_len)
         else:
             self.args = JsonValueMachine(max_len=self.max_args_len, root_object=True)
+
+    def clone(self) -> "ToolCallGrammar":
+        c = object.__new__(ToolCallGrammar)
+        c.schemas = self.schemas            # shared, read-only
+        c.names = self.names
+        c._name_bytes = self._name_bytes
+        c.buf = bytearray(self.buf)
+        c.phase = self.phase
+        c.pos = self.pos
+        c.max_args_len = self.max_args_len
+        c.name_prefix = self.name_prefix    # bytes, immutable
+        c.args = self.args.clone()
+        c.finished = self.finished
+        return c
 
     def allowed_tokens(self) -> Set[int]:
         if self.phase == "pre":

@@ -56,7 +56,7 @@ class Sequence:
         "state", "grammar", "arrival",
     )
 
-    def __init__(self, seq_id: int, request: InferenceRequest):
+    def __init__(self, seq_id: int, request: InferenceRequest, grammar_factory=None):
         self.seq_id = seq_id
         self.request = request
         self.prompt_ids = list(request.prompt_ids)
@@ -66,16 +66,22 @@ class Sequence:
         self.grammar: Optional[ToolCallGrammar] = None
         self.arrival = request.submit_time
         if request.constrained:
-            names = [t["function"]["name"] for t in request.tools]
-            # steer the arguments object to close before the token budget:
-            # the grammar's closing mode needs ~20 tokens of slack for the
-            # name/scaffolding plus the shortest legal unwind
-            max_args = max(16, request.sampling.max_tokens - 24 - max(len(n) for n in names))
-            self.grammar = ToolCallGrammar(
-                tools=request.tools,
-                max_args_len=max_args,
-                pre_in_prompt=request.pre_in_prompt,
-            )
+            if grammar_factory is not None:
+                # BPE path: token-level grammar from the engine's trie
+                self.grammar = grammar_factory(request)
+            else:
+                names = [t["function"]["name"] for t in request.tools]
+                # steer the arguments object to close before the token
+                # budget: the grammar's closing mode needs ~20 tokens of
+                # slack for the name/scaffolding + the shortest unwind
+                max_args = max(
+                    16, request.sampling.max_tokens - 24 - max(len(n) for n in names)
+                )
+                self.grammar = ToolCallGrammar(
+                    tools=request.tools,
+                    max_args_len=max_args,
+                    pre_in_prompt=request.pre_in_prompt,
+                )
 
     @property
     def total_len(self) -> int:
@@ -96,6 +102,9 @@ class Scheduler:
         self.bm = block_manager
         self.device = torch.device(device)
         self.stager = HostStager(device)
+        # set by the engine for BPE tokenizers (token-trie constrained
+        # decoding); None = byte-level ToolCallGrammar
+        self.grammar_factory = None
         self.waiting: List[Sequence] = []
         self.running: List[Sequence] = []   # in admission order (oldest first)
         self._by_id: Dict[int, Sequence] = {}
@@ -124,7 +133,7 @@ class Scheduler:
 
     def add_request(self, request: InferenceRequest) -> Sequence:
         self._seq_counter += 1
-        seq = Sequence(self._seq_counter, request)
+        seq = Sequence(self._seq_counter, request, self.grammar_factory)
         request.seq = seq
         self.waiting.append(seq)
         self._by_id[seq.seq_id] = seq

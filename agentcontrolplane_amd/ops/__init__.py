@@ -106,10 +106,12 @@ def swiglu(gate_up):
     return _impl(gate_up).swiglu(gate_up)
 
 
-def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None, uniforms=None):
-    """Batch sampling: temperature/top-k/top-p with optional per-row boolean
-    vocab masks (constrained decoding) and optional precomputed per-row
-    uniforms (per-request seeded generators).  Returns [B] long."""
+def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None,
+                   uniforms=None, mask_map=None):
+    """Batch sampling: temperature/top-k/top-p with optional boolean vocab
+    masks (constrained decoding; compact [M, V] rows + mask_map [B]
+    indirection for BPE-scale vocabularies) and optional precomputed
+    per-row uniforms (per-request seeded generators).  Returns [B] long."""
     return _impl(logits).softmax_sample(
-        logits, temperatures, top_ks, top_ps, gen, mask, uniforms
+        logits, temperatures, top_ks, top_ps, gen, mask, uniforms, mask_map
     )

@@ -108,11 +108,28 @@ def swiglu(gate_up):
     return out
 
 
-def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None, uniforms=None):
+def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None,
+                   uniforms=None, mask_map=None):
     B, V = logits.shape
     out = torch.empty(B, dtype=torch.long, device=logits.device)
     if uniforms is None:
         uniforms = torch.rand(B, device=logits.device, generator=gen)
+    if V > 512:
+        # BPE-scale vocab: LDS radix-select kernel, bf16 logits unconverted
+        # (csrc/sampling_fullvocab.hip); masks are compact [M, V] rows with
+        # mask_map [B] indirection (-1 = unconstrained row)
+        if logits.dtype not in (torch.bfloat16, torch.float32):
+            logits = logits.float()
+        if mask is not None and mask_map is None:
+            mask_map = torch.arange(B, dtype=torch.int32, device=logits.device)
+        _C.sample_fullvocab(
+            out, logits.contiguous(), temperatures.float().contiguous(),
+            top_ks.long().contiguous(), top_ps.float().contiguous(),
+            uniforms.contiguous(),
+            mask.contiguous() if mask is not None else None,
+            mask_map.contiguous() if mask is not None else None,
+        )
+        return out
     _C.sample(
         out, logits.float().contiguous(), temperatures.float().contiguous(),
         top_ks.long().contiguous(), top_ps.float().contiguous(),

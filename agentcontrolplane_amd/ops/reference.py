@@ -150,14 +150,21 @@ def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
     return (torch.nn.functional.silu(gate) * up).to(gate_up.dtype)
 
 
-def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None, uniforms=None):
+def softmax_sample(logits, temperatures, top_ks, top_ps, gen, mask=None,
+                   uniforms=None, mask_map=None):
     """logits: [B, V] float; temperatures/top_ps: [B] float; top_ks: [B] long
     (0 = off); mask: [B, V] bool (True = allowed) or None; uniforms: [B]
     precomputed draws (per-request seeds) — inverse-CDF like the HIP kernel.
     Greedy when temperature == 0."""
     logits = logits.float().clone()
     B, V = logits.shape
-    if mask is not None:
+    if mask is not None and mask_map is not None:
+        # expand the compact [M, V] mask rows (hip kernel indirection)
+        for i in range(B):
+            mi = int(mask_map[i])
+            if mi >= 0:
+                logits[i].masked_fill_(~mask[mi], float("-inf"))
+    elif mask is not None:
         logits.masked_fill_(~mask, float("-inf"))
     out = torch.empty(B, dtype=torch.long, device=logits.device)
     greedy = temperatures <= 0

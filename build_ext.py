@@ -32,6 +32,7 @@ def build(verbose: bool = False) -> str:
         os.path.join(CSRC, "mfma_probe.hip"),
         os.path.join(CSRC, "gemm_bf16.hip"),
         os.path.join(CSRC, "sampling.hip"),
+        os.path.join(CSRC, "sampling_fullvocab.hip"),
     ]
     module = cpp_extension.load(
         name="_C",

@@ -24,6 +24,9 @@ void launch_decode_attn(void*, const void*, const void*, const void*, const int*
 void launch_prefill_attn(void*, const void*, const void*, const void*, const int*,
                          const int*, const int*, const int*, const int*, const int*,
                          float, int, int, int, int, int, int, int64_t, hipStream_t);
+void launch_sample_fullvocab(int64_t*, const void*, bool, const float*, const int64_t*,
+                             const float*, const float*, const uint8_t*, const int32_t*,
+                             int, int, hipStream_t);
 void launch_sample(int64_t*, const float*, const float*, const int64_t*, const float*,
                    const float*, const uint8_t*, int, int, hipStream_t);
 void launch_prefill_attn_mfma(void*, const void*, const void*, const void*, const int*,
@@ -201,6 +204,33 @@ static void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temps,
                 cur_stream());
 }
 
+static void sample_fullvocab(torch::Tensor out, torch::Tensor logits,
+                             torch::Tensor temps, torch::Tensor top_ks,
+                             torch::Tensor top_ps, torch::Tensor uniforms,
+                             c10::optional<torch::Tensor> mask,
+                             c10::optional<torch::Tensor> mask_map) {
+  CHECK_CUDA(logits); CHECK_CONT(logits);
+  const bool bf16 = logits.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(bf16 || logits.scalar_type() == at::kFloat,
+              "sample_fullvocab wants bf16 or float logits");
+  const int B = logits.size(0), V = logits.size(1);
+  TORCH_CHECK(V <= (1 << 18), "sample_fullvocab supports vocab <= 256k");
+  const uint8_t* mptr = nullptr;
+  const int32_t* mapptr = nullptr;
+  if (mask.has_value()) {
+    TORCH_CHECK(mask->scalar_type() == at::kBool && mask->is_contiguous());
+    TORCH_CHECK(mask_map.has_value(), "compact mask needs mask_map");
+    TORCH_CHECK(mask_map->scalar_type() == at::kInt && mask_map->is_contiguous());
+    TORCH_CHECK(mask_map->numel() == B);
+    mptr = (const uint8_t*)mask->data_ptr();
+    mapptr = mask_map->data_ptr<int32_t>();
+  }
+  launch_sample_fullvocab(out.data_ptr<int64_t>(), logits.data_ptr(), bf16,
+                          temps.data_ptr<float>(), top_ks.data_ptr<int64_t>(),
+                          top_ps.data_ptr<float>(), uniforms.data_ptr<float>(),
+                          mptr, mapptr, B, V, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native runtime: paged-KV block manager + CDNA4 kernels";
   register_block_manager(m);
@@ -214,4 +244,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe);
   m.def("gemm_bf16", &gemm_bf16);
   m.def("sample", &sample);
+  m.def("sample_fullvocab", &sample_fullvocab);
 }

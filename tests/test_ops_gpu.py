@@ -286,3 +286,73 @@ def test_gemm_bf16(m, n, k):
     # bf16 inputs, fp32 accumulation: error scales with sqrt(K)
     tol = 0.03 * math.sqrt(k)  # bf16 output rounding of ~4.5σ results
     assert maxerr(got, want) < tol, maxerr(got, want)
+
+
+@pytest.mark.gpu
+def test_sample_fullvocab_vs_oracle():
+    """csrc/sampling_fullvocab.hip vs the fp32 reference over a Llama-3
+    sized vocabulary: greedy exact, draws match the oracle's inverse CDF
+    for given uniforms, top-k/top-p respect the oracle's kept sets, and
+    compact masks (mask_map indirection) are honored exactly."""
+    from agentcontrolplane_amd.ops import reference
+
+    torch.manual_seed(0)
+    device = "cuda"
+    B, V = 64, 128256
+    gen = torch.Generator(device=device)
+    logits = torch.randn(B, V, device=device, dtype=torch.float32) * 3
+    uniforms = torch.rand(B, device=device)
+    zeros = torch.zeros(B, device=device)
+    ones = torch.ones(B, device=device)
+    offk = torch.zeros(B, dtype=torch.long, device=device)
+
+    # greedy: exact argmax match
+    out = hip().softmax_sample(logits, zeros, offk, ones, gen, uniforms=uniforms)
+    assert torch.equal(out, logits.argmax(-1))
+
+    # temperature-only fp32: draws match the oracle inverse CDF
+    ref = reference.softmax_sample(
+        logits.cpu(), ones.cpu() * 0.8, offk.cpu(), ones.cpu(), None,
+        uniforms=uniforms.cpu())
+    out = hip().softmax_sample(logits, ones * 0.8, offk, ones, gen,
+                               uniforms=uniforms)
+    frac = (out.cpu() == ref).float().mean().item()
+    assert frac >= 0.95, f"only {frac:.2%} of draws matched the oracle"
+
+    # top-k: every draw inside the oracle's kept set {x >= kth}
+    k = 50
+    out = hip().softmax_sample(logits, ones, torch.full_like(offk, k), ones,
+                               gen, uniforms=uniforms)
+    kth = torch.topk(logits, k, dim=-1).values[:, -1]
+    picked = logits[torch.arange(B, device=device), out]
+    assert (picked >= kth - 1e-6).all()
+
+    # top-p: draws inside the oracle nucleus
+    p = 0.7
+    out = hip().softmax_sample(logits, ones, offk, ones * p, gen,
+                               uniforms=uniforms)
+    probs = torch.softmax(logits, -1)
+    sp, si = torch.sort(probs, descending=True, dim=-1)
+    cum = sp.cumsum(-1)
+    for i in range(B):
+        keep_n = int((cum[i] - sp[i] <= p).sum())
+        kept = set(si[i, :max(1, keep_n)].tolist())
+        assert int(out[i]) in kept
+
+    # bf16 logits path + compact mask rows with indirection
+    lb = logits.bfloat16()
+    n_masked = 8
+    mask = torch.zeros(n_masked, V, dtype=torch.bool, device=device)
+    allowed_ids = torch.randint(0, V, (n_masked, 37), device=device)
+    for j in range(n_masked):
+        mask[j, allowed_ids[j]] = True
+    mask_map = torch.full((B,), -1, dtype=torch.int32, device=device)
+    mask_map[: n_masked] = torch.arange(n_masked, dtype=torch.int32, device=device)
+    out = hip().softmax_sample(lb, ones, offk, ones, gen, mask,
+                               uniforms=uniforms, mask_map=mask_map)
+    for j in range(n_masked):
+        assert bool(mask[j, out[j]]), "masked row sampled a disallowed token"
+    # unmasked rows unaffected by the compact mask
+    out_ref = hip().softmax_sample(lb, ones, offk, ones, gen,
+                                   uniforms=uniforms)
+    assert torch.equal(out[n_masked:], out_ref[n_masked:])
