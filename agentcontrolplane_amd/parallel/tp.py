@@ -88,11 +88,129 @@ def shard_from_full(full, shard, rank: int, world: int) -> None:
         shard.layers.append(sl)
 
 
+_META_GROUP = None
+_META_INIT = False
+
+
+def _meta_group():
+    """A gloo side-group for the metadata broadcast: the step wire is
+    consumed on the host (workers rebuild host lists + device tensors), so
+    a CPU-tensor broadcast avoids the GPU round-trip a NCCL broadcast
+    would force — and tensor broadcast replaces round 1's per-step
+    ``broadcast_object_list`` pickling."""
+    global _META_GROUP, _META_INIT
+    if not _META_INIT:
+        _META_INIT = True
+        if dist.get_backend() != "gloo":
+            _META_GROUP = dist.new_group(backend="gloo")
+    return _META_GROUP
+
+
+def _pack_wire(wire: Optional[dict]) -> torch.Tensor:
+    """Flatten the step metadata into one int64 tensor (layout mirrored by
+    _unpack_wire).  op codes: 0 = step, 1 = stop."""
+    if wire is None or wire.get("op") == "stop":
+        return torch.tensor([1], dtype=torch.int64)
+    out = [0]
+    tok = wire["token_ids"]
+    out.append(len(tok))
+    out.extend(tok)
+    out.extend(wire["positions"])
+    out.extend(wire["slot_mapping"])
+    out.append(wire["num_prefill_tokens"])
+    prefills = wire["prefills"]
+    out.append(len(prefills))
+    for (sid, q, sl, c, bt, nl) in prefills:
+        out.extend((sid, q, sl, c, 1 if nl else 0, len(bt)))
+        out.extend(bt)
+    dec = wire["decode_seq_ids"]
+    out.append(len(dec))
+    out.extend(dec)
+    tables = wire["decode_block_tables"]
+    if tables is None:
+        out.append(-1)
+    else:
+        width = len(tables[0]) if tables else 0
+        out.append(width)
+        for row in tables:
+            out.extend(row)
+    lens = wire["decode_seq_lens"]
+    out.append(-1 if lens is None else len(lens))
+    if lens is not None:
+        out.extend(lens)
+    lr = wire["logit_rows"]
+    out.append(len(lr))
+    out.extend(lr)
+    ss = wire["sample_seq_ids"]
+    out.append(len(ss))
+    out.extend(ss)
+    return torch.tensor(out, dtype=torch.int64)
+
+
+def _unpack_wire(t: torch.Tensor) -> Optional[dict]:
+    data = t.tolist()
+    pos = 0
+
+    def take(n):
+        nonlocal pos
+        v = data[pos:pos + n]
+        pos += n
+        return v
+
+    def one():
+        return take(1)[0]
+
+    if one() == 1:
+        return STOP
+    n = one()
+    token_ids = take(n)
+    positions = take(n)
+    slot_mapping = take(n)
+    num_prefill_tokens = one()
+    prefills = []
+    for _ in range(one()):
+        sid, q, sl, c, nl, btn = take(6)
+        prefills.append((sid, q, sl, c, take(btn), bool(nl)))
+    n_dec = one()
+    decode_seq_ids = take(n_dec)
+    width = one()
+    tables = None
+    if width >= 0:
+        tables = [take(width) for _ in range(n_dec)]
+    n_lens = one()
+    lens = take(n_lens) if n_lens >= 0 else None
+    logit_rows = take(one())
+    sample_seq_ids = take(one())
+    return {
+        "token_ids": token_ids,
+        "positions": positions,
+        "slot_mapping": slot_mapping,
+        "prefills": prefills,
+        "num_prefill_tokens": num_prefill_tokens,
+        "decode_seq_ids": decode_seq_ids,
+        "decode_block_tables": tables,
+        "decode_seq_lens": lens,
+        "logit_rows": logit_rows,
+        "sample_seq_ids": sample_seq_ids,
+    }
+
+
 def broadcast_step(wire: Optional[dict], src: int = 0, group=None) -> Optional[dict]:
-    """Rank 0 sends the step metadata (or STOP); workers receive it."""
-    box = [wire]
-    dist.broadcast_object_list(box, src=src, group=group)
-    return box[0]
+    """Rank 0 sends the step metadata (or STOP); workers receive it.  One
+    size broadcast + one packed int64 payload broadcast over the gloo
+    side-group — no pickling, no GPU round-trip."""
+    g = group if group is not None else _meta_group()
+    if dist.get_rank() == src:
+        payload = _pack_wire(wire)
+        size = torch.tensor([payload.numel()], dtype=torch.int64)
+        dist.broadcast(size, src=src, group=g)
+        dist.broadcast(payload, src=src, group=g)
+        return wire
+    size = torch.tensor([0], dtype=torch.int64)
+    dist.broadcast(size, src=src, group=g)
+    payload = torch.empty(int(size[0]), dtype=torch.int64)
+    dist.broadcast(payload, src=src, group=g)
+    return _unpack_wire(payload)
 
 
 def run_tp_worker(engine) -> None:
