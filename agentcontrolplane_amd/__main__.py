@@ -32,16 +32,22 @@ def cmd_serve(args) -> None:
 
         device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
         model = args.model if device == "cuda" else "tiny"
-        from .engine.engine import InferenceEngine
-
-        engine = InferenceEngine(
-            EngineConfig(
-                model=model,
-                device=device,
-                checkpoint_path=args.checkpoint,
-                num_kv_blocks=args.kv_blocks,
-            )
+        ecfg = EngineConfig(
+            model=model,
+            device=device,
+            checkpoint_path=args.checkpoint,
+            num_kv_blocks=args.kv_blocks,
         )
+        n_gpus = getattr(args, "gpus", 1) or 1
+        if device.startswith("cuda") and n_gpus > 1:
+            # DP request routing across this node's GPUs (parallel/router)
+            from .parallel.router import EnginePool
+
+            engine = EnginePool.build(ecfg, n_gpus)
+        else:
+            from .engine.engine import InferenceEngine
+
+            engine = InferenceEngine(ecfg)
     store = None
     if getattr(args, "kube", False):
         # apiserver-backed store: in-cluster credentials or --kube-url
@@ -198,6 +204,9 @@ def main() -> None:
     s.add_argument("--checkpoint", default=None, help="safetensors checkpoint dir")
     s.add_argument("--device", default=None, help="cuda | cpu | none (no engine)")
     s.add_argument("--kv-blocks", type=int, default=None)
+    s.add_argument("--gpus", type=int, default=1,
+                   help="DP-route requests across N GPUs of this node "
+                   "(one engine per GPU, least-loaded routing)")
     s.add_argument("--auto-approve", action="store_true")
     s.add_argument("--kube", action="store_true",
                    help="back the store with the Kubernetes apiserver "
