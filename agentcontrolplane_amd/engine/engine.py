@@ -407,8 +407,15 @@ class InferenceEngine:
             try:
                 if r.error is not None:
                     callback(None, r.error)
-                else:
-                    callback(self._result_of(r, t0), None)
+                    return
+                try:
+                    result = self._result_of(r, t0)
+                except Exception as e:  # result extraction failed: the
+                    # request must still complete (a swallowed exception
+                    # here left the Task hanging to its timeout)
+                    callback(None, e)
+                    return
+                callback(result, None)
             except Exception:  # noqa: BLE001 — callback bugs must not kill the loop
                 import traceback
 

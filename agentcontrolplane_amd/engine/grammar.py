@@ -123,7 +123,11 @@ class JsonValueMachine:
                 opts |= _bs(b"{[")
             return opts
         if s == "string":
-            return _bs(STRING_SAFE, b'"', b"\\", UTF8_LEADS)
+            opts = _bs(STRING_SAFE, b'"', b"\\")
+            if self.count + 4 < self.max_len:
+                # a codepoint's continuations must fit inside the budget
+                opts |= _bs(UTF8_LEADS)
+            return opts
         if s.startswith("str_c"):
             _, lo, hi = s.split(":")
             return _bs(range(int(lo), int(hi) + 1))
@@ -179,6 +183,11 @@ class JsonValueMachine:
         """Over the length budget: only bytes on the shortest path to an
         accepting state, so generation always terminates in valid JSON."""
         s = self.state
+        if s.startswith("str_c"):
+            # mid-codepoint: the only legal path to acceptance FINISHES the
+            # UTF-8 sequence (truncating it would leave an undecodable buf)
+            _, lo, hi = s.split(":")
+            return _bs(range(int(lo), int(hi) + 1))
         if s in ("string", "obj_key_str"):
             return {0x22}  # '"'
         if s in ("str_escape", "obj_key_escape"):

@@ -161,3 +161,31 @@ def test_schema_keys_and_tool_names_with_json_specials():
         assert n == name
         parsed = json.loads(args)
         assert set(parsed) == set(props)
+
+
+def test_utf8_lead_near_budget_always_closes_valid():
+    """Regression (r02 GPU bench): budget exhaustion right after a UTF-8
+    lead byte must FINISH the codepoint on the closing path — an empty
+    allowed set here let the sampler emit byte 0 mid-sequence and
+    parse() crashed on invalid UTF-8."""
+    import random
+
+    for seed in range(300):
+        rng = random.Random(seed)
+        g = ToolCallGrammar(tool_names=["t"], max_args_len=rng.randint(8, 24))
+        steps = 0
+        while not g.finished and steps < 600:
+            allowed = g.allowed_tokens()
+            assert allowed, f"empty allowed set in phase {g.phase} (seed {seed})"
+            # bias toward UTF-8 leads so budget exhaustion lands mid-codepoint
+            leads = [b for b in allowed if 0xC2 <= b < 0xF5]
+            tok = rng.choice(leads) if leads and rng.random() < 0.6 else rng.choice(
+                sorted(allowed)
+            )
+            g.advance(tok)
+            steps += 1
+            if g.phase == "done":
+                g.advance(EOT)
+        assert g.finished, f"seed {seed} never finished"
+        name, args = g.parse()  # must decode + json-parse cleanly
+        assert name == "t"
