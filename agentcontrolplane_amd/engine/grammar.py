@@ -266,7 +266,10 @@ class JsonValueMachine:
             self.count -= 1
             return self.advance(b)
         if s == "string":
-            if c == b'"':
+            n, lo, hi = _utf8_first_range(b)
+            if n:
+                self.state = f"str_c{n}:{lo}:{hi}"
+            elif c == b'"':
                 self._value_done()
             elif c == b"\\":
                 self.state = "str_escape"

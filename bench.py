@@ -151,7 +151,7 @@ def main() -> None:
     p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree (else DP sharding)")
     p.add_argument("--device", default=None)
     p.add_argument("--kv-blocks", type=int, default=None)
-    p.add_argument("--prefill-budget", type=int, default=8192)
+    p.add_argument("--prefill-budget", type=int, default=16384)
     p.add_argument(
         "--sub-agents", type=int, default=0,
         help="N child agents under the bench agent (BASELINE.json config 4: "

@@ -170,9 +170,12 @@ def test_utf8_lead_near_budget_always_closes_valid():
     parse() crashed on invalid UTF-8."""
     import random
 
-    for seed in range(300):
+    # tight budgets exercise the closing path; roomy budgets exercise
+    # long string values where lead->continuation bookkeeping must hold
+    budgets = [lambda r: r.randint(8, 24)] * 300 + [lambda r: 100] * 100
+    for seed, budget in enumerate(budgets):
         rng = random.Random(seed)
-        g = ToolCallGrammar(tool_names=["t"], max_args_len=rng.randint(8, 24))
+        g = ToolCallGrammar(tool_names=["t"], max_args_len=budget(rng))
         steps = 0
         while not g.finished and steps < 600:
             allowed = g.allowed_tokens()
