@@ -134,6 +134,9 @@ class EngineConfig:
     # syncs (hides per-step host time); auto-disabled for TP and whenever a
     # grammar-constrained decode is running
     async_scheduling: bool = True
+    # fold forced grammar byte runs (MID scaffolding, schema literals) into
+    # one prefill chunk instead of sequential masked decode steps
+    grammar_fold: bool = True
     tensor_parallel: int = 1
     seed: int = 0
     request_timeout_s: float = 600.0

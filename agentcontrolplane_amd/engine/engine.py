@@ -154,6 +154,7 @@ class InferenceEngine:
             self._mask_rows = 128
             self._ps_mask = stager.persistent(self._mask_rows, (self._live,), _np.bool_)
             self._ps_maskmap = stager.persistent(cap, (), _np.int32)
+        self._spec_bail = False  # one-shot: set by a fold-pending commit
         self._lock = threading.Lock()
         self._work = threading.Condition(self._lock)
         self._pending: List[InferenceRequest] = []
@@ -457,12 +458,17 @@ class InferenceEngine:
                     self._m["busy_time_s"] += time.monotonic() - t0
                     continue
                 # a step is in flight: launch its successor speculatively,
-                # then sync/commit the pending one
-                nxt = self._schedule_and_launch(pending["batch"]) if spec_ok else None
+                # then sync/commit the pending one.  A detected forced run
+                # requests ONE spec bail so the following commit can fold
+                # it into a prefill chunk (no in-flight successor to
+                # invalidate).
+                do_spec = spec_ok and not self._spec_bail
+                self._spec_bail = False
+                nxt = self._schedule_and_launch(pending["batch"]) if do_spec else None
                 if nxt is not None:
                     self._m["spec_steps"] = self._m.get("spec_steps", 0) + 1
                 t2 = time.monotonic()
-                self._commit(pending)
+                self._commit(pending, has_successor=nxt is not None)
                 self._m["sample_time_s"] += time.monotonic() - t2
                 if nxt is not None and nxt.get("deferred"):
                     # grammar-carrying speculative step: its forward is
@@ -719,7 +725,7 @@ class InferenceEngine:
         batch._tokens_gpu = tokens  # speculative successors gather from this
         return {"batch": batch, "seqs": seqs, "tokens": tokens}
 
-    def _commit(self, pending) -> None:
+    def _commit(self, pending, has_successor: bool = False) -> None:
         """Sync the sampled tokens and advance sequence/grammar state."""
         seqs = pending["seqs"]
         if not seqs:
@@ -748,6 +754,23 @@ class InferenceEngine:
                     # never walk past the RoPE table (cfg.max_position)
                     self._m["requests_completed"] += 1
                     self.scheduler.finish_seq(s, "length")
+                    continue
+                # forced-run folding: a run of singleton-allowed bytes (MID
+                # scaffolding, schema literals, a lone tool name's tail)
+                # becomes one chunked-prefill extension instead of one
+                # masked decode step per byte
+                if self.cfg.grammar_fold and s.state == "decode":
+                    forced = s.grammar.forced_run()
+                    if len(forced) >= 4 and (
+                        s.total_len + len(forced) < self._ctx_limit - 1
+                    ):
+                        if has_successor:
+                            # the in-flight speculative step would sample a
+                            # stale position after a fold: bail spec once so
+                            # the NEXT commit folds with nothing in flight
+                            self._spec_bail = True
+                        else:
+                            self._fold_forced(s, forced)
                 continue
             if tok == self._eot:
                 s.request.output_ids.pop()
@@ -764,6 +787,30 @@ class InferenceEngine:
                 # per-sequence list folds into the prompt on preempt)
                 self._m["requests_completed"] += 1
                 self.scheduler.finish_seq(s, "length")
+
+    def _fold_forced(self, s, forced) -> None:
+        """Fold a forced byte run into the sequence's prompt so the next
+        step prefills it as ONE chunk (with its following logits row)
+        instead of len(forced) sequential masked decode steps.
+
+        Bookkeeping mirrors recompute-preemption's prompt fold
+        (scheduler._preempt) minus the KV free: the KV-resident prefix
+        stays (= bm.seq_len: every token except the just-sampled one),
+        the folded bytes count as generated output, and the grammar
+        advances through them now so the next sampled token sees the
+        post-scaffolding state."""
+        for b in forced:
+            s.request.output_ids.append(b)
+            s.grammar.advance(b)
+            self._m["generated_tokens"] += 1
+        s.prompt_ids = s.prompt_ids + s.output_ids + list(forced)
+        s.output_ids = []
+        s.num_processed = self.bm.seq_len(s.seq_id)
+        s.state = "prefill"
+        self._m["grammar_folds"] = self._m.get("grammar_folds", 0) + 1
+        self._m["grammar_folded_tokens"] = (
+            self._m.get("grammar_folded_tokens", 0) + len(forced)
+        )
 
     # ------------------------------------------------------------- metrics
 

@@ -609,6 +609,39 @@ class ToolCallGrammar:
         c.finished = self.finished
         return c
 
+    def forced_run(self, cap: int = 64) -> List[int]:
+        """The upcoming run of bytes with exactly one legal choice — the
+        scaffolding the engine can fold into one prefill chunk instead of
+        one masked decode step per byte (MID is 16 forced bytes; schema
+        literals and a single matching tool name add more).  Guarded by
+        cheap phase checks so unconstrained positions cost nothing; EOT
+        never appears in a forced run."""
+        # fast bail: states that always have >1 legal byte
+        if self.phase in ("done",):
+            return []
+        if self.phase == "args":
+            a = self.args
+            if isinstance(a, JsonValueMachine):
+                return []  # free JSON: almost never singleton for long
+            cur = a._cur()
+            if cur is None or cur[0] != "lit":
+                return []
+        out: List[int] = []
+        g = None
+        while len(out) < cap:
+            src = g if g is not None else self
+            al = src.allowed_tokens()
+            if len(al) != 1:
+                break
+            b = next(iter(al))
+            if b == EOT or b > 255:
+                break
+            if g is None:
+                g = self.clone()
+            g.advance(b)
+            out.append(b)
+        return out
+
     def allowed_tokens(self) -> Set[int]:
         if self.phase == "pre":
             return {self.PRE[self.pos]}

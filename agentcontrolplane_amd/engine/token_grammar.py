@@ -170,6 +170,12 @@ class TokenGrammar:
 
         return EOT
 
+    def forced_run(self, cap: int = 64):
+        """No byte-level folding for BPE: multi-byte tokens already cover
+        forced scaffolding in a handful of steps, and folding would need
+        token-boundary alignment of the forced byte string."""
+        return []
+
     @property
     def phase(self) -> str:
         return self.byte_grammar.phase
