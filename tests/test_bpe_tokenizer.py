@@ -238,3 +238,73 @@ def test_checkpoint_dir_with_tokenizer_json(bpe_path, tmp_path):
         json.loads(res.tool_calls[0]["function"]["arguments"])
     finally:
         eng.stop()
+
+
+@pytest.mark.gpu
+def test_bpe_serving_on_gpu_fullvocab_sampler():
+    """The whole BPE serving path on hardware: HF tokenizer + token-trie
+    masks + the radix-select full-vocab sampler (live vocab > 512 routes
+    to csrc/sampling_fullvocab.hip) through the engine on cuda:0,
+    producing a grammar-valid executable tool call."""
+    import dataclasses
+
+    from tokenizers import Tokenizer, models, pre_tokenizers, decoders, trainers
+
+    from agentcontrolplane_amd.engine.config import PRESETS
+
+    tok = Tokenizer(models.BPE(unk_token=None))
+    tok.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+    tok.decoder = decoders.ByteLevel()
+    trainer = trainers.BpeTrainer(
+        vocab_size=2000, special_tokens=[], show_progress=False,
+        initial_alphabet=pre_tokenizers.ByteLevel.alphabet(),
+    )
+    corpus = ['{"name": "calc__add", "arguments": {"a": 1, "b": 2}}',
+              "the quick brown fox jumps over the lazy dog"] * 80
+    tok.train_from_iterator(corpus, trainer)
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as td:
+        path = f"{td}/tokenizer.json"
+        tok.save(path)
+        PRESETS["tiny-gpu-bpe"] = dataclasses.replace(
+            PRESETS["tiny-gpu"], name="tiny-gpu-bpe", vocab_size=4096
+        )
+        try:
+            eng = InferenceEngine(
+                EngineConfig(model="tiny-gpu-bpe", device="cuda",
+                             num_kv_blocks=256, tokenizer_path=path,
+                             max_prefill_tokens=512),
+                start=True,
+            )
+            try:
+                assert eng._live > 512  # full-vocab sampler route
+                assert eng.scheduler.grammar_factory is not None
+                tools = [{"type": "function", "function": {
+                    "name": "calc__add",
+                    "parameters": {"type": "object",
+                                   "properties": {"a": {"type": "number"},
+                                                  "b": {"type": "number"}},
+                                   "required": ["a", "b"]}}}]
+                res = eng.chat(
+                    [{"role": "user", "content": "add 1 and 2"}],
+                    tools=tools,
+                    sampling=SamplingParams(max_tokens=64, temperature=0.9,
+                                            top_p=0.95, top_k=50,
+                                            tool_choice="required"),
+                )
+                assert res.finish_reason == "tool_calls"
+                call = res.tool_calls[0]["function"]
+                assert call["name"] == "calc__add"
+                args = json.loads(call["arguments"])
+                assert set(args) == {"a", "b"}
+                # free-text turn exercises the unmasked full-vocab draw
+                res2 = eng.chat(
+                    [{"role": "user", "content": "hello"}],
+                    sampling=SamplingParams(max_tokens=8, temperature=0.8),
+                )
+                assert res2.completion_tokens <= 8
+            finally:
+                eng.stop()
+        finally:
+            PRESETS.pop("tiny-gpu-bpe", None)
