@@ -257,11 +257,16 @@ def test_bpe_serving_on_gpu_fullvocab_sampler():
     tok.decoder = decoders.ByteLevel()
     trainer = trainers.BpeTrainer(
         vocab_size=2000, special_tokens=[], show_progress=False,
+        min_frequency=2,
         initial_alphabet=pre_tokenizers.ByteLevel.alphabet(),
     )
-    corpus = ['{"name": "calc__add", "arguments": {"a": 1, "b": 2}}',
-              "the quick brown fox jumps over the lazy dog"] * 80
+    # merge-rich corpus: hundreds of distinct repeated words so the vocab
+    # grows well past 512 (routing to the full-vocab sampler)
+    corpus = ['{"name": "calc__add", "arguments": {"a": 1, "b": 2}}'] * 40 + [
+        " ".join(f"tok{i:04d}" for i in range(j, j + 40)) for j in range(0, 800, 10)
+    ] * 4
     tok.train_from_iterator(corpus, trainer)
+    assert tok.get_vocab_size() > 512, tok.get_vocab_size()
     import tempfile
 
     with tempfile.TemporaryDirectory() as td:
