@@ -96,8 +96,9 @@ def validate(obj, schema, path="$"):
 
 
 class MockAPIServer:
-    def __init__(self):
+    def __init__(self, port: int = 0):
         self.store = ResourceStore()
+        self._port = port
         self.schemas = _load_crd_schemas()
         self._gen = [0]
         outer = self
@@ -261,7 +262,7 @@ class MockAPIServer:
                 else:
                     self._status(404, "NotFound", m["name"])
 
-        self._srv = ThreadingHTTPServer(("127.0.0.1", 0), Handler)
+        self._srv = ThreadingHTTPServer(("127.0.0.1", self._port), Handler)
         self._thread = threading.Thread(target=self._srv.serve_forever, daemon=True)
         self._thread.start()
         self.base = f"http://127.0.0.1:{self._srv.server_address[1]}"
@@ -275,3 +276,5 @@ class MockAPIServer:
 
     def shutdown(self):
         self._srv.shutdown()
+        self._srv.server_close()  # release the listening socket (restarts
+        # re-bind the same port in the reconnect test)

@@ -266,3 +266,58 @@ def test_leader_election_failover():
     wait_for(lambda: (shared.get("LLM", "l2") or {}).get("status", {}).get("ready"),
              timeout=10)
     b.stop()
+
+
+def test_watch_survives_apiserver_restart():
+    """The per-kind watch threads re-list and reconnect after the
+    apiserver drops (KubeStore._watch_loop's 410/error path) — events for
+    objects created after the restart still arrive."""
+    import queue as _q
+    import time
+
+    srv = MockAPIServer()
+    port = srv._srv.server_address[1]
+    ks = KubeStore(base_url=srv.base)
+    try:
+        q = ks.watch(kinds={"Task"})
+        ks.create({
+            "apiVersion": "acp.humanlayer.dev/v1alpha1", "kind": "Task",
+            "metadata": {"name": "before"},
+            "spec": {"agentRef": {"name": "a"}, "userMessage": "x"},
+        })
+        ev = q.get(timeout=5)
+        assert ev.obj["metadata"]["name"] == "before"
+        # kill the apiserver; bring a fresh one up on the SAME port
+        srv.shutdown()
+        time.sleep(0.3)
+        srv2 = MockAPIServer(port=port)
+        try:
+            # wait for the store's reconnect loop to re-establish, then
+            # a new object's ADDED must flow through the same queue
+            deadline = time.monotonic() + 15
+            created = False
+            seen = False
+            while time.monotonic() < deadline and not seen:
+                if not created:
+                    try:
+                        ks.create({
+                            "apiVersion": "acp.humanlayer.dev/v1alpha1",
+                            "kind": "Task", "metadata": {"name": "after"},
+                            "spec": {"agentRef": {"name": "a"},
+                                     "userMessage": "y"},
+                        })
+                        created = True
+                    except Exception:
+                        time.sleep(0.2)  # server not up yet
+                        continue
+                try:
+                    ev = q.get(timeout=1.0)
+                except _q.Empty:
+                    continue
+                if ev.obj.get("metadata", {}).get("name") == "after":
+                    seen = True
+            assert seen, "watch did not recover after apiserver restart"
+        finally:
+            srv2.shutdown()
+    finally:
+        ks.close()
