@@ -90,6 +90,10 @@ class KubeStore:
         self._watch_threads: List[threading.Thread] = []
         self._stop = False
         self._watch_kinds = watch_kinds or (list(ACP_KINDS) + [EVENT])
+        # single-namespace deployment scope (the manager's POD_NAMESPACE,
+        # like the reference's namespaced kustomize install); cluster-wide
+        # informers would need the un-namespaced list/watch paths
+        self.namespace = os.environ.get("POD_NAMESPACE", "default")
 
     # ------------------------------------------------------------------ paths
 
@@ -164,7 +168,7 @@ class KubeStore:
         if label_selector:
             params["labelSelector"] = ",".join(f"{k}={v}" for k, v in label_selector.items())
         if namespace is None:
-            namespace = "default"  # single-namespace deployments; parity with local store default
+            namespace = self.namespace  # single-namespace deployment scope
         r = self._client.get(self._path(kind, namespace), params=params)
         if r.status_code == 404:
             return []
@@ -239,7 +243,7 @@ class KubeStore:
     def _watch_loop(self, kind: str) -> None:
         import httpx
 
-        ns = "default"
+        ns = self.namespace
         rv = None
         while not self._stop:
             try:
