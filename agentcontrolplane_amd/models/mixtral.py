@@ -58,11 +58,15 @@ class MixtralForCausalLM(LlamaForCausalLM):
             self.expert_gate_up.append(randw(self.num_experts, 2 * self.inter, h))
             self.expert_down.append(randw(self.num_experts, h, self.inter))
 
-    #: below this many tokens the MoE layer is weight-read-bound, so the
-    #: dense path (ALL experts via one bmm pair) costs the same wall time as
-    #: the sparse loop, launches 3 kernels instead of ~2·E, and has static
-    #: control flow — hipGraph-capturable
-    dense_moe_threshold = 160
+    #: below this many tokens the dense path (ALL experts via one bmm
+    #: pair, 3 launches, static control flow — hipGraph-capturable) beats
+    #: the sparse loop (~5·E launches of small ragged GEMMs): under ~160
+    #: tokens the layer is weight-read-bound either way, and between 160
+    #: and ~1k the sparse loop's tiny per-expert GEMMs run far below peak
+    #: while its launch storm dominates the host (r02 Mixtral profile:
+    #: 28% GPU busy, MT16x16/MT32x32 GEMM + index kernels all over the
+    #: top-20).  4x FLOPs of dense only start losing past ~1k tokens.
+    dense_moe_threshold = 1024
 
     def _moe_mlp(self, li: int, x: torch.Tensor) -> torch.Tensor:
         """x: [N, H] → [N, H] via top-k expert mixture."""
