@@ -60,13 +60,17 @@ class MixtralForCausalLM(LlamaForCausalLM):
 
     #: below this many tokens the dense path (ALL experts via one bmm
     #: pair, 3 launches, static control flow — hipGraph-capturable) beats
-    #: the sparse loop (~5·E launches of small ragged GEMMs): under ~160
-    #: tokens the layer is weight-read-bound either way, and between 160
-    #: and ~1k the sparse loop's tiny per-expert GEMMs run far below peak
-    #: while its launch storm dominates the host (r02 Mixtral profile:
-    #: 28% GPU busy, MT16x16/MT32x32 GEMM + index kernels all over the
-    #: top-20).  4x FLOPs of dense only start losing past ~1k tokens.
-    dense_moe_threshold = 1024
+    #: the sparse loop (~5·E launches of small ragged GEMMs).  The r02
+    #: Mixtral profile shows the sparse loop leaves the GPU 28% busy at
+    #: 64-task serving (MT16x16/MT32x32 GEMMs + index kernels swamp the
+    #: host), and pricing says dense wins through ~1k tokens — but
+    #: raising this to 1024 faulted on hardware during the warmup wave
+    #: (GPU memory access fault; suspected: the stride-0-expanded bmm /
+    #: 470 MB transpose-contiguous at N≈1k, never exercised before), so
+    #: it stays at the proven 160 pending a dense-path rewrite without
+    #: the expand+transpose (docs/roadmap.md: grouped GEMM is the real
+    #: fix for the mid-N launch storm).
+    dense_moe_threshold = 160
 
     def _moe_mlp(self, li: int, x: torch.Tensor) -> torch.Tensor:
         """x: [N, H] → [N, H] via top-k expert mixture."""
