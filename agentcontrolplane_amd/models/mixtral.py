@@ -11,6 +11,8 @@ through an RCCL all-to-all over xGMI.
 """
 from __future__ import annotations
 
+import os
+
 from typing import List
 
 import torch
@@ -70,7 +72,7 @@ class MixtralForCausalLM(LlamaForCausalLM):
     #: it stays at the proven 160 pending a dense-path rewrite without
     #: the expand+transpose (docs/roadmap.md: grouped GEMM is the real
     #: fix for the mid-N launch storm).
-    dense_moe_threshold = 160
+    dense_moe_threshold = int(os.environ.get("ACP_MOE_DENSE_THRESHOLD", "160"))
 
     def _moe_mlp(self, li: int, x: torch.Tensor) -> torch.Tensor:
         """x: [N, H] → [N, H] via top-k expert mixture."""
@@ -100,20 +102,31 @@ class MixtralForCausalLM(LlamaForCausalLM):
         return out
 
     def _moe_dense(self, li: int, x: torch.Tensor, topi, topw) -> torch.Tensor:
-        """All-experts bmm + routed combine (same numerics as the sparse
-        loop: non-selected experts get weight 0)."""
+        """All-experts batched GEMMs + routed combine (same numerics as the
+        sparse loop: non-selected experts get weight 0).
+
+        Formulated [E, N, *] end-to-end: the weights provide the batch dim
+        (op_B transposes handled inside rocBLAS) and the activations stay
+        in row-major [E, N, C] throughout — no stride-0 expanded operand
+        and no transpose-contiguous round trips (the earlier
+        bmm-[E,2I,N]-then-transpose formulation copied two [E, N, 2I]
+        intermediates per layer, and its expanded operand faulted on
+        hardware when admitted at N≈1k)."""
         E = self.num_experts
         N, H = x.shape
-        # [E, 2I, H] @ [E, H, N] -> [E, 2I, N]
-        xb = x.t().unsqueeze(0).expand(E, H, N)
-        gate_up = torch.bmm(self.expert_gate_up[li], xb)
-        act = ops.swiglu(gate_up.transpose(1, 2).contiguous())       # [E, N, I]
-        y = torch.bmm(self.expert_down[li], act.transpose(1, 2))     # [E, H, N]
+        # [E, N, H] @ [E, H, 2I] -> [E, N, 2I]; the small explicit replica
+        # of x (E·N·H, ~64 MB at N=1k) buys plain strided batched GEMMs —
+        # no stride-0 operands anywhere
+        xe = x.unsqueeze(0).expand(E, N, H).contiguous()
+        gate_up = torch.matmul(xe, self.expert_gate_up[li].transpose(1, 2))
+        act = ops.swiglu(gate_up.contiguous())                       # [E, N, I]
+        # [E, N, I] @ [E, I, H] -> [E, N, H]
+        y = torch.matmul(act, self.expert_down[li].transpose(1, 2))
         # routing weights as a dense [N, E] matrix
         w = torch.zeros(N, E, dtype=x.dtype, device=x.device)
         w.scatter_(1, topi, topw.to(x.dtype))
-        # out[n, h] = sum_e w[n, e] * y[e, h, n]
-        return torch.einsum("ne,ehn->nh", w, y)
+        # out[n, h] = sum_e w[n, e] * y[e, n, h]
+        return torch.einsum("ne,enh->nh", w, y)
 
     def forward(self, batch) -> torch.Tensor:
         cfg = self.cfg
