@@ -60,18 +60,17 @@ class MixtralForCausalLM(LlamaForCausalLM):
             self.expert_gate_up.append(randw(self.num_experts, 2 * self.inter, h))
             self.expert_down.append(randw(self.num_experts, h, self.inter))
 
-    #: below this many tokens the dense path (ALL experts via one bmm
-    #: pair, 3 launches, static control flow — hipGraph-capturable) beats
-    #: the sparse loop (~5·E launches of small ragged GEMMs).  The r02
-    #: Mixtral profile shows the sparse loop leaves the GPU 28% busy at
-    #: 64-task serving (MT16x16/MT32x32 GEMMs + index kernels swamp the
-    #: host), and pricing says dense wins through ~1k tokens — but
-    #: raising this to 1024 faulted on hardware during the warmup wave
-    #: (GPU memory access fault; suspected: the stride-0-expanded bmm /
-    #: 470 MB transpose-contiguous at N≈1k, never exercised before), so
-    #: it stays at the proven 160 pending a dense-path rewrite without
-    #: the expand+transpose (docs/roadmap.md: grouped GEMM is the real
-    #: fix for the mid-N launch storm).
+    #: below this many tokens the dense path (ALL experts via batched
+    #: GEMMs, static control flow — hipGraph-capturable) beats the sparse
+    #: loop (~5·E launches of small ragged GEMMs).  The r02 Mixtral
+    #: profile prices dense as the win through ~1k tokens (sparse leaves
+    #: the GPU 28% busy at 64-task serving), but admitting N≈1k FAULTS on
+    #: hardware under BOTH dense formulations tried (the old stride-0
+    #: expand+transpose AND the plain strided [E,N,*] rewrite) — the
+    #: fault is suspected inside the batched-GEMM library at
+    #: [8, ~1k, 28672]-class shapes and needs a standalone repro before
+    #: this threshold moves (ACP_MOE_DENSE_THRESHOLD overrides for
+    #: experiments; docs/roadmap.md tracks the grouped-GEMM fix).
     dense_moe_threshold = int(os.environ.get("ACP_MOE_DENSE_THRESHOLD", "160"))
 
     def _moe_mlp(self, li: int, x: torch.Tensor) -> torch.Tensor:
