@@ -5,11 +5,16 @@ langchaingo_client.go): ``chat()`` turns an ACP context window into token
 ids, submits an InferenceRequest, and blocks on its future while the engine
 thread continuously batches every in-flight request into shared GPU steps.
 
-Sampling is over the tokenizer's live vocabulary (bytes + specials); the
-full-vocab LM-head GEMM is still computed (that cost is real and stays in
-the measured path), the slice only bounds the sampling distribution to
-decodable ids.  Constrained (tool-call) sequences additionally mask to the
-grammar's legal next bytes each step.
+Sampling is over the tokenizer's live vocabulary — bytes + specials for
+the synthetic byte tokenizer, the full BPE vocabulary when a
+tokenizer.json is loaded (the ≤512-entry bitonic sampler vs the
+radix-select full-vocab kernel route on that size); the full-vocab
+LM-head GEMM is always computed (that cost is real and stays in the
+measured path).  Constrained (tool-call) sequences additionally mask to
+the grammar's legal next tokens each step — single bytes under the byte
+PDA, whole tokens under the token-trie walk (token_grammar.py) — and
+runs of forced scaffolding bytes fold into one prefill chunk
+(_fold_forced) instead of sequential masked decode steps.
 """
 from __future__ import annotations
 
