@@ -58,10 +58,15 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
 
   // LDS: K row-major [32][128] with XOR swizzle on 16-B units to break the
   // ds_read_b128 16-way conflict (guide G4: byte ^= (row&7)<<4); V
-  // transposed [128][32] (row = d, 64 B) read conflict-light.  Double
-  // buffered: tile t+1 stages while tile t computes (one barrier/tile).
+  // transposed [128][32+2] — the 2-element row pad makes the row stride
+  // 17 words (coprime with the 32 banks), so the B-fragment b128 reads
+  // (one V row per lane) start on all-distinct banks instead of the
+  // 2-bank/16-way pileup a 64-B stride gives, and the 2-B transpose
+  // scatter drops from 16-way to ~4-way (r02 PMC: this kernel's LDS
+  // conflict cycles matched its MFMA-busy cycles before the pad).
+  // Double buffered: tile t+1 stages while tile t computes.
   __shared__ bf16_t Kt[2][KVBLK][D_HEAD];
-  __shared__ bf16_t Vt[2][D_HEAD][KVBLK];
+  __shared__ bf16_t Vt[2][D_HEAD][KVBLK + 2];
   __shared__ float bcast[4][QBLK];
   __shared__ int bt[512];
   const int nblk = (S + kv_block - 1) / kv_block;
