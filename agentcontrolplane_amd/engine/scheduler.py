@@ -224,6 +224,14 @@ class Scheduler:
                 cur = self.bm.seq_len(s.seq_id)
                 have = len(self.bm.block_table(s.seq_id))
                 chunk = fit_chunk(cur, have, want)
+                if chunk <= 0 and want > 0 and self.retired:
+                    # the continuation cache must never starve live work:
+                    # reclaim retirees for a running prefill exactly as the
+                    # admission path does for waiting sequences
+                    before = self.bm.free_blocks
+                    self._reclaim(self.bm.free_blocks + (want + bs - 1) // bs)
+                    avail += self.bm.free_blocks - before
+                    chunk = fit_chunk(cur, have, want)
                 if chunk > 0:
                     prefills.append((s, chunk))
                     budget -= chunk
@@ -275,6 +283,22 @@ class Scheduler:
         if spec_after is None:
             decode_seqs = [s for s in self.running if s.state == DECODE]
         if not prefills and not decode_seqs:
+            # starvation break: running PREFILL sequences exist but none
+            # could take a chunk (pool exhausted mid-prefill).  Decode
+            # sequences are preempted above, but two-plus concurrent
+            # prefills can jointly exhaust the pool with no decode victim
+            # — admission bounds each SINGLE sequence to fit the pool, so
+            # preempting the youngest stalled prefill (recompute-style)
+            # guarantees the oldest eventually completes.  Folding makes
+            # this reachable in practice (folded sequences re-enter
+            # prefill holding their KV).
+            if spec_after is None:
+                stalled = [s for s in self.running if s.state == PREFILL]
+                if len(stalled) > 1 and budget > 0:
+                    victim = stalled[-1]  # youngest
+                    self.running.remove(victim)
+                    self._preempt(victim)
+                    preempted.append(victim)
             return SchedulerOutput(batch=None, preempted=preempted)
 
         # 3) materialize the flat batch

@@ -348,3 +348,45 @@ def test_grammar_fold_produces_identical_tool_calls():
     assert m_plain.get("grammar_folds", 0) == 0
     # folding removed sequential steps
     assert m_fold["steps"] < m_plain["steps"]
+
+
+def test_grammar_fold_under_preemption_pressure():
+    """Folding + recompute preemption compose: a tiny KV pool forces
+    preemptions while constrained turns fold scaffolding into their
+    prompts; every tool call must still parse and execute."""
+    import json as _json
+
+    from agentcontrolplane_amd.engine.request import SamplingParams
+
+    eng = InferenceEngine(
+        EngineConfig(model="tiny", device="cpu", num_kv_blocks=48,
+                     kv_block_size=16, max_prefill_tokens=128, seed=11),
+        start=True,
+    )
+    try:
+        tools = [{"type": "function", "function": {
+            "name": "t__go",
+            "parameters": {"type": "object",
+                           "properties": {"msg": {"type": "string"}},
+                           "required": ["msg"]}}}]
+        reqs = []
+        for i in range(6):
+            reqs.append(eng.chat_async(
+                [{"role": "user", "content": f"task {i} " + "x" * 200}],
+                tools,
+                SamplingParams(max_tokens=40, temperature=0.9,
+                               tool_choice="required"),
+                lambda r, e: None,
+            ))
+        for r in reqs:
+            r.wait(60)
+        m = eng.metrics()
+        assert m.get("grammar_folds", 0) >= 1
+        for r in reqs:
+            assert r.finish_reason == "tool_calls", r.finish_reason
+            assert r.seq.grammar.finished
+            name, args = r.seq.grammar.parse()
+            assert name == "t__go"
+            assert isinstance(_json.loads(args)["msg"], str)
+    finally:
+        eng.stop()
