@@ -313,3 +313,25 @@ def test_bpe_serving_on_gpu_fullvocab_sampler():
                 eng.stop()
         finally:
             PRESETS.pop("tiny-gpu-bpe", None)
+
+
+def test_bpe_chat_stream_deltas(bpe_path):
+    """chat_stream with a BPE tokenizer: deltas arrive per decoded UTF-8
+    span via token_bytes (round 1 assumed one token = one byte)."""
+    eng = InferenceEngine(
+        EngineConfig(model="tiny", device="cpu", num_kv_blocks=512,
+                     tokenizer_path=bpe_path, max_prefill_tokens=512),
+        start=True,
+    )
+    try:
+        deltas, finals = [], []
+        for kind, payload in eng.chat_stream(
+            [{"role": "user", "content": "stream some text"}],
+            sampling=SamplingParams(max_tokens=16, temperature=1.0),
+        ):
+            (deltas if kind == "delta" else finals).append(payload)
+        assert finals and finals[0].completion_tokens <= 16
+        joined = "".join(deltas)
+        assert joined == finals[0].text  # deltas reassemble the final text
+    finally:
+        eng.stop()
