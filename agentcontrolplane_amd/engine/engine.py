@@ -30,7 +30,7 @@ from .config import EngineConfig
 from .kv import make_block_manager
 from .request import ChatResult, InferenceRequest, SamplingParams
 from .scheduler import Scheduler, Sequence
-from .tokenizer import EOT, N_SPECIAL, TOOL_CALL_START, ByteTokenizer
+from .tokenizer import N_SPECIAL, ByteTokenizer
 
 
 class InferenceEngine:

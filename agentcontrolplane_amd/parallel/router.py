@@ -21,7 +21,7 @@ the component the reference lacks a counterpart for.
 from __future__ import annotations
 
 import threading
-from typing import List, Optional
+from typing import List
 
 
 class EnginePool:

@@ -12,7 +12,6 @@ from __future__ import annotations
 import json
 import os
 import threading
-import time
 from typing import List, Optional
 
 from .tracer import Span, Tracer
