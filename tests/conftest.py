@@ -19,7 +19,7 @@ def store():
     s.close()
 
 
-def wait_for(predicate, timeout=10.0, interval=0.01):
+def wait_for(predicate, timeout=25.0, interval=0.01):
     """Poll until predicate() is truthy; return its value or raise."""
     import time
 

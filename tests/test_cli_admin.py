@@ -141,7 +141,7 @@ def test_serve_subprocess_boots_and_runs_loop(tmp_path):
         stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
     )
     try:
-        client = httpx.Client(base_url=f"http://127.0.0.1:{port}", timeout=5)
+        client = httpx.Client(base_url=f"http://127.0.0.1:{port}", timeout=20)
         deadline = time.time() + 30
         while time.time() < deadline:
             try:

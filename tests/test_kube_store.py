@@ -248,7 +248,7 @@ def test_leader_election_failover():
         cp.LEADER_LEASE_DURATION = 1.0
         cp.LEADER_RETRY = 0.1
     a.start(leader_elect=True)
-    wait_for(lambda: a.is_leader, timeout=5)
+    wait_for(lambda: a.is_leader, timeout=15)
     b.start(leader_elect=True)
     time.sleep(0.5)
     assert a.is_leader and not b.is_leader
