@@ -272,3 +272,62 @@ def test_approval_gate_e2e_through_http(hl_server):
         assert any(m.get("role") == "tool" for m in task["status"]["contextWindow"])
     finally:
         cp.stop()
+
+
+def test_rejection_feeds_back_through_http(hl_server):
+    """A rejected approval must NOT fail the ToolCall: the reference marks
+    it Succeeded with result "Rejected: <comment>" so the refusal feeds
+    back into the next LLM turn (toolcall/state_machine.go:153-160) —
+    here through the real HTTP client against the mock server."""
+    hl_server.auto = "reject"
+    from agentcontrolplane_amd.llmclient.mock import MockLLMClient
+    from agentcontrolplane_amd.llmclient.factory import LLMClientFactory
+    from agentcontrolplane_amd.runtime import ControlPlane
+
+    factory = HTTPHumanLayerClientFactory(api_base=hl_server.base,
+                                          api_key="hl-key")
+    cp = ControlPlane(
+        humanlayer_factory=factory,
+        llm_client_factory=LLMClientFactory(mock_factory=lambda llm: MockLLMClient()),
+        llm_probe=False,
+    )
+    cp.start()
+    try:
+        cp.store.create(make_resource(SECRET, "hl-secret",
+                                      spec={"data": {"k": "hl-key"}},
+                                      api_version="v1"))
+        cp.store.create(make_resource(CONTACT_CHANNEL, "approvals", spec={
+            "type": "slack",
+            "apiKeyFrom": {"secretKeyRef": {"name": "hl-secret", "key": "k"}},
+            "slack": {"channelOrUserID": "CREJECT"},
+        }))
+        cp.store.create(make_resource(LLM, "mock-llm", spec={"provider": "mock"}))
+        cp.mcp.register_inproc("tools", {"add": lambda a=0, b=0, **_: str(a + b)})
+        cp.store.create(make_resource(MCP_SERVER, "tools", spec={
+            "transport": "inproc",
+            "approvalContactChannel": {"name": "approvals"},
+        }))
+        cp.store.create(make_resource(AGENT, "rej-agent", spec={
+            "llmRef": {"name": "mock-llm"}, "system": "sys",
+            "mcpServers": [{"name": "tools"}],
+        }))
+        wait_for(lambda: (cp.store.get(AGENT, "rej-agent") or {})
+                 .get("status", {}).get("ready"), timeout=15)
+        cp.store.create(make_resource(TASK, "rej-task", spec={
+            "agentRef": {"name": "rej-agent"},
+            "userMessage": "use the add tool",
+        }))
+        task = wait_for(
+            lambda: (cp.store.get(TASK, "rej-task") or {}).get("status", {})
+            .get("phase") == TaskPhase.FINAL_ANSWER
+            and cp.store.get(TASK, "rej-task"),
+            timeout=30,
+        )
+        # the rejection reached the context window as a tool result
+        tool_msgs = [m for m in task["status"]["contextWindow"]
+                     if m.get("role") == "tool"]
+        assert tool_msgs and tool_msgs[0]["content"].startswith("Rejected")
+        fc = next(iter(hl_server.calls.values()))
+        assert fc["status"]["approved"] is False
+    finally:
+        cp.stop()
