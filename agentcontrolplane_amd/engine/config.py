@@ -24,6 +24,8 @@ class ModelConfig:
     rms_eps: float = 1e-5
     max_position: int = 8192
     tie_embeddings: bool = False
+    # Qwen2-family attention bias on the QKV projection
+    attention_bias: bool = False
     # MoE (Mixtral): 0 experts = dense
     num_experts: int = 0
     num_experts_per_tok: int = 2
@@ -48,6 +50,24 @@ class ModelConfig:
 
 PRESETS = {
     "llama3-8b": ModelConfig(),
+    # Llama-2 family: same architecture, smaller rope base / context / vocab
+    "llama2-7b": ModelConfig(
+        name="llama2-7b", vocab_size=32000, hidden_size=4096,
+        intermediate_size=11008, num_layers=32, num_heads=32,
+        num_kv_heads=32, rope_theta=10000.0, rms_eps=1e-5, max_position=4096,
+    ),
+    "llama2-13b": ModelConfig(
+        name="llama2-13b", vocab_size=32000, hidden_size=5120,
+        intermediate_size=13824, num_layers=40, num_heads=40,
+        num_kv_heads=40, rope_theta=10000.0, max_position=4096,
+    ),
+    # Qwen2: GQA + QKV bias (the one architectural delta from Llama)
+    "qwen2-7b": ModelConfig(
+        name="qwen2-7b", vocab_size=152064, hidden_size=3584,
+        intermediate_size=18944, num_layers=28, num_heads=28,
+        num_kv_heads=4, rope_theta=1000000.0, rms_eps=1e-6,
+        max_position=8192, attention_bias=True,
+    ),
     "llama3-70b": ModelConfig(
         name="llama3-70b",
         hidden_size=8192,
@@ -83,6 +103,20 @@ PRESETS = {
         dtype="bfloat16",
     ),
     # tiny configs for CPU tests
+    # tiny bias variant (Qwen2-shaped attention for CPU tests)
+    "tiny-bias": ModelConfig(
+        name="tiny-bias",
+        vocab_size=512,
+        hidden_size=64,
+        intermediate_size=128,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=16,
+        max_position=4096,
+        attention_bias=True,
+        dtype="float32",
+    ),
     "tiny": ModelConfig(
         name="tiny",
         vocab_size=512,

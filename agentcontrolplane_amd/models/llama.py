@@ -36,7 +36,11 @@ def _dtype_of(cfg: ModelConfig):
 
 
 class LlamaLayerWeights:
-    __slots__ = ("input_norm", "qkv", "o", "post_norm", "gate_up", "down")
+    __slots__ = ("input_norm", "qkv", "qkv_bias", "o", "post_norm", "gate_up", "down")
+
+    def __init__(self):
+        # Qwen2-style attention bias; None for Llama/Mixtral
+        self.qkv_bias = None
 
 
 class LlamaForCausalLM:
@@ -107,6 +111,10 @@ class LlamaForCausalLM:
             lw = LlamaLayerWeights()
             lw.input_norm = torch.ones(h, dtype=self.dtype, device=self.device)
             lw.qkv = randw(qd + 2 * kvd, h)
+            if self.cfg.attention_bias:
+                b = torch.empty(qd + 2 * kvd, dtype=self.dtype, device=self.device)
+                b.normal_(0.0, std, generator=g)
+                lw.qkv_bias = b
             lw.o = randw(h, qd)
             lw.post_norm = torch.ones(h, dtype=self.dtype, device=self.device)
             lw.gate_up = randw(2 * self.inter, h)
@@ -173,6 +181,8 @@ class LlamaForCausalLM:
             else:
                 normed, residual = ops.fused_add_rmsnorm(x, residual, lw.input_norm, cfg.rms_eps)
             qkv = ops.linear(normed, lw.qkv)
+            if lw.qkv_bias is not None:
+                qkv = qkv + lw.qkv_bias
             # strided views into the fused GEMM output: the HIP kernels take
             # a token stride, so no .contiguous() copies on the hot path
             row = qkv.stride(0)

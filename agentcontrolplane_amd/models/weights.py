@@ -36,6 +36,12 @@ def _llama_name_map(model) -> Dict[str, torch.Tensor]:
             qd // model.tp_world : (qd + kvd) // model.tp_world
         ]
         out[p + "self_attn.v_proj.weight"] = lw.qkv[(qd + kvd) // model.tp_world :]
+        if lw.qkv_bias is not None:
+            out[p + "self_attn.q_proj.bias"] = lw.qkv_bias[: qd // model.tp_world]
+            out[p + "self_attn.k_proj.bias"] = lw.qkv_bias[
+                qd // model.tp_world : (qd + kvd) // model.tp_world
+            ]
+            out[p + "self_attn.v_proj.bias"] = lw.qkv_bias[(qd + kvd) // model.tp_world :]
         out[p + "self_attn.o_proj.weight"] = lw.o
         if lw.gate_up is not None:
             inter = lw.gate_up.shape[0] // 2
@@ -142,6 +148,17 @@ def load_checkpoint(model, path: str) -> None:
                 vw[r * kv_per * hd : (r + 1) * kv_per * hd],
             ]
         ).to(dev, dt)
+        if cfg.attention_bias:
+            qb = get(p + "self_attn.q_proj.bias")
+            kb = get(p + "self_attn.k_proj.bias")
+            vb = get(p + "self_attn.v_proj.bias")
+            lw.qkv_bias = torch.cat(
+                [
+                    qb[r * q_per * hd : (r + 1) * q_per * hd],
+                    kb[r * kv_per * hd : (r + 1) * kv_per * hd],
+                    vb[r * kv_per * hd : (r + 1) * kv_per * hd],
+                ]
+            ).to(dev, dt)
         ow = get(p + "self_attn.o_proj.weight")
         lw.o = ow[:, r * q_per * hd : (r + 1) * q_per * hd].to(dev, dt)
         if not is_moe:

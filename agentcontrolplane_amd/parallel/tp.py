@@ -73,6 +73,19 @@ def shard_from_full(full, shard, rank: int, world: int) -> None:
                 ]
             )
         )
+        if getattr(fl, "qkv_bias", None) is not None:
+            qb = fl.qkv_bias[: nq * hd]
+            kb = fl.qkv_bias[nq * hd : (nq + nkv) * hd]
+            vb = fl.qkv_bias[(nq + nkv) * hd :]
+            sl.qkv_bias = to_dev(
+                torch.cat(
+                    [
+                        qb[rank * q_per * hd : (rank + 1) * q_per * hd],
+                        kb[rank * kv_per * hd : (rank + 1) * kv_per * hd],
+                        vb[rank * kv_per * hd : (rank + 1) * kv_per * hd],
+                    ]
+                )
+            )
         sl.o = to_dev(fl.o[:, rank * q_per * hd : (rank + 1) * q_per * hd])
         gate = fl.gate_up[: cfg.intermediate_size]
         up = fl.gate_up[cfg.intermediate_size :]

@@ -390,3 +390,50 @@ def test_grammar_fold_under_preemption_pressure():
             assert isinstance(_json.loads(args)["msg"], str)
     finally:
         eng.stop()
+
+
+def test_attention_bias_model_family():
+    """Qwen2-family QKV bias: the tiny-bias preset serves end-to-end, and
+    save/load round-trips the bias through the HF naming
+    (self_attn.{q,k,v}_proj.bias)."""
+    import tempfile
+
+    import torch
+
+    from agentcontrolplane_amd.engine.request import SamplingParams
+    from agentcontrolplane_amd.models import create_model
+    from agentcontrolplane_amd.engine.config import PRESETS
+    from agentcontrolplane_amd.models.weights import load_checkpoint, save_checkpoint
+
+    eng = InferenceEngine(
+        EngineConfig(model="tiny-bias", device="cpu", num_kv_blocks=256),
+        start=True,
+    )
+    try:
+        assert eng.model.layers[0].qkv_bias is not None
+        res = eng.chat(
+            [{"role": "user", "content": "hello"}],
+            sampling=SamplingParams(max_tokens=8, temperature=0.8),
+        )
+        assert res.completion_tokens <= 8
+    finally:
+        eng.stop()
+
+    # checkpoint round-trip incl. bias
+    m = create_model(PRESETS["tiny-bias"], EngineConfig(model="tiny-bias", device="cpu"), "cpu")
+    m.random_init(5)
+    with tempfile.TemporaryDirectory() as td:
+        save_checkpoint(m, td)
+        m2 = create_model(PRESETS["tiny-bias"], EngineConfig(model="tiny-bias", device="cpu"), "cpu")
+        load_checkpoint(m2, td)
+        assert torch.equal(m.layers[0].qkv_bias, m2.layers[0].qkv_bias)
+        assert torch.equal(m.layers[1].qkv, m2.layers[1].qkv)
+
+
+def test_llama2_presets_exist():
+    from agentcontrolplane_amd.engine.config import PRESETS
+
+    for name in ("llama2-7b", "llama2-13b", "qwen2-7b"):
+        cfg = PRESETS[name]
+        assert cfg.params_bytes() > 1 << 30  # real-sized
+    assert PRESETS["qwen2-7b"].attention_bias
