@@ -102,6 +102,20 @@ PRESETS = {
         max_position=2048,
         dtype="bfloat16",
     ),
+    # tiny GPU config with Qwen2-style attention bias
+    "tiny-gpu-bias": ModelConfig(
+        name="tiny-gpu-bias",
+        vocab_size=512,
+        hidden_size=256,
+        intermediate_size=512,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        max_position=2048,
+        attention_bias=True,
+        dtype="bfloat16",
+    ),
     # tiny configs for CPU tests
     # tiny bias variant (Qwen2-shaped attention for CPU tests)
     "tiny-bias": ModelConfig(

@@ -412,3 +412,27 @@ def test_tp_shard_path_math_on_gpu():
         out = sh.forward(mk()).float()
         assert out.shape == ref_logits.shape
         assert torch.isfinite(out).all()
+
+
+@pytest.mark.gpu
+def test_attention_bias_serving_on_gpu():
+    """Qwen2-style QKV bias through the CDNA4 kernel path: the bias add
+    precedes the strided q/k/v views, so the fused-QKV stride handling
+    must hold on the result tensor."""
+    from agentcontrolplane_amd.engine.config import EngineConfig
+    from agentcontrolplane_amd.engine.engine import InferenceEngine
+    from agentcontrolplane_amd.engine.request import SamplingParams
+
+    eng = InferenceEngine(
+        EngineConfig(model="tiny-gpu-bias", device="cuda", num_kv_blocks=128),
+        start=True,
+    )
+    try:
+        assert eng.model.layers[0].qkv_bias is not None
+        res = eng.chat(
+            [{"role": "user", "content": "bias check"}],
+            sampling=SamplingParams(max_tokens=8, temperature=0.8),
+        )
+        assert res.completion_tokens <= 8
+    finally:
+        eng.stop()
